@@ -1,0 +1,170 @@
+#!/usr/bin/env python
+"""Benchmark harness — the driver contract.
+
+Measures the BASELINE.json headline metric: train frames/sec (whole node) on
+the BAIR 64x64 / vgg_64 / seq_len=30 config, bf16 compute, synthetic data
+(random-init weights; no network for datasets). A "step" is one full training
+iteration: forward over the 30-step recurrence, both backward phases, all five
+Adam steps — nothing is skipped inside the timed region.
+
+Launch: python bench.py --gpus N --steps K --warmup W
+  N>1 is launched by the driver via torch.distributed.run (one rank per GPU,
+  RCCL over xGMI); this script reads RANK/LOCAL_RANK/WORLD_SIZE from the env.
+
+Prints ONE JSON line from rank 0.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+from p2pvg_amd.core import Config
+from p2pvg_amd.models import P2PModel
+
+
+def make_synthetic_batch(cfg, device, seed: int):
+    """Synthetic BAIR-shaped clip batch: (T, B, C, H, W) in [0,1], smooth in t."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    t, b = cfg.max_seq_len, cfg.batch_size
+    c, s = cfg.channels, cfg.image_width
+    base = torch.rand(1, b, c, s, s, generator=g)
+    drift = torch.randn(t, b, c, 1, 1, generator=g) * 0.05
+    x = (base + drift.cumsum(0)).clamp_(0, 1)
+    return x.to(device)
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=32, help="per-GPU batch size")
+    p.add_argument("--seq_len", type=int, default=30)
+    p.add_argument("--backbone", type=str, default="vgg")
+    p.add_argument("--image_width", type=int, default=64)
+    p.add_argument("--dataset", type=str, default="bair")
+    p.add_argument("--dtype", type=str, default="bf16")
+    p.add_argument("--kernels", type=str, default="auto")
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+
+    use_cuda = torch.cuda.is_available()
+    if args.kernels != "auto":
+        os.environ["P2PVG_KERNELS"] = args.kernels
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if use_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
+
+    cfg = Config(
+        dataset=args.dataset,
+        backbone=args.backbone,
+        image_width=args.image_width,
+        channels=3 if args.dataset in ("bair", "weizmann") else 1,
+        batch_size=args.batch,
+        max_seq_len=args.seq_len,
+        skip_prob=0.0,       # fixed-length timing: every step does full work
+        weight_align=0.5,
+        weight_cpc=100.0,
+        dtype=args.dtype if use_cuda else "fp32",
+        device=str(device),
+        ddp=world > 1,
+    )
+    if args.dataset == "h36m":
+        cfg.backbone = "mlp"
+
+    torch.manual_seed(1234 + rank)
+    np.random.seed(1234 + rank)
+
+    model = P2PModel(cfg).to(device)
+    if world > 1:
+        from p2pvg_amd.parallel import DDPGradSync
+
+        sync = DDPGradSync(model, bucket_mb=cfg.bucket_mb)
+        sync.broadcast_parameters()
+
+    x = make_synthetic_batch(cfg, device, seed=1234 + rank)
+    amp = cfg.dtype == "bf16" and use_cuda
+
+    def one_step():
+        model.zero_grad(set_to_none=False)
+        if amp:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                model(x, 0, len(x) - 1)
+        else:
+            model(x, 0, len(x) - 1)
+
+    # warmup
+    for _ in range(args.warmup):
+        one_step()
+
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else None)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    global_batch = args.batch * world
+    frames = global_batch * args.seq_len * args.steps
+    fps = frames / elapsed
+
+    if rank == 0:
+        result = {
+            "metric": "train_frames_per_sec",
+            "value": fps,
+            "unit": "frames/s",
+            "n_gpus": world if use_cuda else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": cfg.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": f"p2pvg_{args.backbone}_{args.image_width}",
+                "dataset": args.dataset,
+                "global_batch": global_batch,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

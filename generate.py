@@ -1,0 +1,155 @@
+#!/usr/bin/env python
+"""Generation CLI — reference-compatible entrypoint (reference generate.py).
+
+Loads a checkpoint (ours or a reference-written .pth), rebuilds the model from
+the checkpoint's own config, reads an input video (.mp4/.gif via PIL, or a
+directory of frames, or --start_img/--end_img image pair — the reference
+declared that path but never defined the flags, generate.py:93-96), and
+generates lengths [10, 20, 30] x 5 samples, saving PNG rows and GIFs.
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import random
+
+import numpy as np
+import torch
+
+from p2pvg_amd.models import P2PModel
+from p2pvg_amd.utils import config_from_states, load_checkpoint
+from p2pvg_amd.utils.image import make_grid, save_gif, save_image, to_uint8_hwc
+from p2pvg_amd.utils.vis import add_gt_cp_border, add_samples_cp_border
+
+
+def read_video(vid_name: str) -> torch.Tensor:
+    """Read a video file or frame directory -> (t, 1, c, h, w) in [0,1]."""
+    frames = []
+    if os.path.isdir(vid_name):
+        from PIL import Image
+
+        for f in sorted(os.listdir(vid_name)):
+            if f.lower().endswith((".png", ".jpg", ".jpeg")):
+                with Image.open(os.path.join(vid_name, f)) as im:
+                    frames.append(np.asarray(im.convert("RGB"), dtype=np.float32) / 255.0)
+    else:
+        from PIL import Image, ImageSequence
+
+        with Image.open(vid_name) as im:
+            for frame in ImageSequence.Iterator(im):
+                frames.append(np.asarray(frame.convert("RGB"), dtype=np.float32) / 255.0)
+    if not frames:
+        raise ValueError(f"no frames read from {vid_name}")
+    t = torch.from_numpy(np.stack(frames)).permute(0, 3, 1, 2)
+    return t.unsqueeze(1)
+
+
+def read_image_pair(start_img: str, end_img: str) -> torch.Tensor:
+    from PIL import Image
+
+    out = []
+    for p in (start_img, end_img):
+        with Image.open(p) as im:
+            out.append(
+                torch.from_numpy(
+                    np.asarray(im.convert("RGB"), dtype=np.float32) / 255.0
+                ).permute(2, 0, 1)
+            )
+    return torch.stack(out).unsqueeze(1)
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--ckpt", type=str, required=True, help="model .pth file")
+    parser.add_argument("--video", type=str, default="", help=".mp4/.gif video or frame dir")
+    parser.add_argument("--start_img", type=str, default="")
+    parser.add_argument("--end_img", type=str, default="")
+    parser.add_argument("--output_root", type=str, default="gen_outputs")
+    parser.add_argument("--seed", type=int, default=1)
+    parser.add_argument("--device", type=str, default="auto")
+    args = parser.parse_args()
+
+    states = load_checkpoint(args.ckpt)
+    cfg = config_from_states(states)
+    cfg.batch_size = 1
+    cfg.device = args.device
+
+    random.seed(args.seed)
+    np.random.seed(args.seed)
+    torch.manual_seed(args.seed)
+    if torch.cuda.is_available():
+        torch.cuda.manual_seed_all(args.seed)
+
+    device = torch.device(cfg.resolved_device())
+    model = P2PModel(cfg).to(device)
+    model.load(states=states)
+    model.eval()
+
+    nsamples, ndisplays = 5, 5
+    gen_lengths = [10, 20, 30]
+
+    if args.video:
+        seq = read_video(args.video)
+    elif args.start_img:
+        assert args.end_img, "--start_img requires --end_img"
+        seq = read_image_pair(args.start_img, args.end_img)
+    else:
+        raise SystemExit("provide --video or --start_img/--end_img")
+
+    if cfg.channels == 1 and seq.shape[2] == 3:
+        seq = seq.mean(dim=2, keepdim=True)
+    seq = seq.to(device)
+    seq_len = len(seq)
+
+    os.makedirs(args.output_root, exist_ok=True)
+
+    with torch.no_grad():
+        for length_to_gen in gen_lengths:
+            output_cp_ix = length_to_gen - 1
+            samples = []
+            for _ in range(nsamples):
+                out = model.p2p_generate(seq, length_to_gen, output_cp_ix, model_mode="full")
+                samples.append(torch.stack(out))
+            samples = torch.stack(samples)
+
+            idx = np.random.choice(len(samples), ndisplays, replace=False)
+            samples_to_save = samples[idx].cpu().float()
+
+            padded_seq = seq.clone().cpu().float()
+            x_cp = padded_seq[seq_len - 1]
+            if length_to_gen > seq_len:
+                pad = x_cp.unsqueeze(0).repeat(length_to_gen - seq_len, 1, 1, 1, 1)
+                padded_seq = torch.cat([padded_seq, pad], dim=0)
+
+            seq_b = add_gt_cp_border(padded_seq, seq_len, length_to_gen)
+            samples_to_save = add_samples_cp_border(samples_to_save, seq_len, length_to_gen)
+
+            save_image(
+                make_grid(seq_b[:, 0], nrow=len(seq_b), padding=0),
+                os.path.join(args.output_root, f"len_{length_to_gen}-gt.png"),
+            )
+
+            block = []
+            for ix, s in enumerate(samples_to_save):
+                row = make_grid(s[:, 0], nrow=len(s), padding=0)
+                save_image(row, os.path.join(args.output_root, f"len_{length_to_gen}-gen_{ix:03d}.png"))
+                block.append(row)
+            save_image(
+                torch.cat(block, 1),
+                os.path.join(args.output_root, f"len_{length_to_gen}-gen_full.png"),
+            )
+
+            for ix, s in enumerate(samples_to_save):
+                frames = [to_uint8_hwc(s[t, 0]) for t in range(len(s))]
+                save_gif(os.path.join(args.output_root, f"len_{length_to_gen}-gen_{ix:03d}.gif"), frames)
+
+            gifs = [
+                to_uint8_hwc(make_grid(samples_to_save[:, t, 0], nrow=ndisplays, padding=0))
+                for t in range(length_to_gen)
+            ]
+            save_gif(os.path.join(args.output_root, f"len_{length_to_gen}-gen_full.gif"), gifs)
+            print(f"[*] wrote length-{length_to_gen} outputs to {args.output_root}")
+
+
+if __name__ == "__main__":
+    main()

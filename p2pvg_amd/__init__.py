@@ -1,0 +1,11 @@
+"""p2pvg_amd — MI355X-native point-to-point video generation framework.
+
+A from-scratch framework with the capabilities of yccyenchicheng/p2pvg
+(ICCV 2019, arXiv:1904.02912), built MI355X-first: PyTorch-ROCm driver,
+hand-written CDNA4 (gfx950) HIP kernels for the hot ops, hipGraph-captured
+recurrence, RCCL-over-xGMI data parallelism.
+"""
+
+__version__ = "0.1.0"
+
+from .core import Config  # noqa: F401

@@ -1,0 +1,137 @@
+"""VGG-style backbones for 64x64 and 128x128 frames.
+
+Capability parity with reference models/vgg_64.py:16-105 and
+models/vgg_128.py:16-120 (shape tables in SURVEY §2.3): 3x3-conv blocks with
+2x2 maxpool between encoder stages; decoder is nearest-upsample + conv blocks
+with channel-concat skips. state_dict keys match the reference modules.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from .blocks import vgg_layer
+
+
+class Encoder64(nn.Module):
+    def __init__(self, dim: int, nc: int = 1):
+        super().__init__()
+        self.dim = dim
+        self.c1 = nn.Sequential(vgg_layer(nc, 64), vgg_layer(64, 64))
+        self.c2 = nn.Sequential(vgg_layer(64, 128), vgg_layer(128, 128))
+        self.c3 = nn.Sequential(
+            vgg_layer(128, 256), vgg_layer(256, 256), vgg_layer(256, 256)
+        )
+        self.c4 = nn.Sequential(
+            vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+        )
+        self.c5 = nn.Sequential(
+            nn.Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
+        )
+        self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
+
+    def forward(self, x):
+        h1 = self.c1(x)
+        h2 = self.c2(self.mp(h1))
+        h3 = self.c3(self.mp(h2))
+        h4 = self.c4(self.mp(h3))
+        h5 = self.c5(self.mp(h4))
+        return h5.view(-1, self.dim), [h1, h2, h3, h4]
+
+
+class Decoder64(nn.Module):
+    def __init__(self, dim: int, nc: int = 1):
+        super().__init__()
+        self.dim = dim
+        self.upc1 = nn.Sequential(
+            nn.ConvTranspose2d(dim, 512, 4, 1, 0),
+            nn.BatchNorm2d(512),
+            nn.LeakyReLU(0.2, inplace=True),
+        )
+        self.upc2 = nn.Sequential(
+            vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 256)
+        )
+        self.upc3 = nn.Sequential(
+            vgg_layer(256 * 2, 256), vgg_layer(256, 256), vgg_layer(256, 128)
+        )
+        self.upc4 = nn.Sequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
+        self.upc5 = nn.Sequential(
+            vgg_layer(64 * 2, 64),
+            nn.ConvTranspose2d(64, nc, 3, 1, 1),
+            nn.Sigmoid(),
+        )
+        self.up = nn.UpsamplingNearest2d(scale_factor=2)
+
+    def forward(self, inp):
+        vec, skip = inp
+        d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
+        d2 = self.upc2(torch.cat([self.up(d1), skip[3]], 1))
+        d3 = self.upc3(torch.cat([self.up(d2), skip[2]], 1))
+        d4 = self.upc4(torch.cat([self.up(d3), skip[1]], 1))
+        return self.upc5(torch.cat([self.up(d4), skip[0]], 1))
+
+
+class Encoder128(nn.Module):
+    def __init__(self, dim: int, nc: int = 1):
+        super().__init__()
+        self.dim = dim
+        self.c1 = nn.Sequential(vgg_layer(nc, 64), vgg_layer(64, 64))
+        self.c2 = nn.Sequential(vgg_layer(64, 128), vgg_layer(128, 128))
+        self.c3 = nn.Sequential(
+            vgg_layer(128, 256), vgg_layer(256, 256), vgg_layer(256, 256)
+        )
+        self.c4 = nn.Sequential(
+            vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+        )
+        self.c5 = nn.Sequential(
+            vgg_layer(512, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+        )
+        self.c6 = nn.Sequential(
+            nn.Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
+        )
+        self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
+
+    def forward(self, x):
+        h1 = self.c1(x)
+        h2 = self.c2(self.mp(h1))
+        h3 = self.c3(self.mp(h2))
+        h4 = self.c4(self.mp(h3))
+        h5 = self.c5(self.mp(h4))
+        h6 = self.c6(self.mp(h5))
+        return h6.view(-1, self.dim), [h1, h2, h3, h4, h5]
+
+
+class Decoder128(nn.Module):
+    def __init__(self, dim: int, nc: int = 1):
+        super().__init__()
+        self.dim = dim
+        self.upc1 = nn.Sequential(
+            nn.ConvTranspose2d(dim, 512, 4, 1, 0),
+            nn.BatchNorm2d(512),
+            nn.LeakyReLU(0.2, inplace=True),
+        )
+        self.upc2 = nn.Sequential(
+            vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+        )
+        self.upc3 = nn.Sequential(
+            vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 256)
+        )
+        self.upc4 = nn.Sequential(
+            vgg_layer(256 * 2, 256), vgg_layer(256, 256), vgg_layer(256, 128)
+        )
+        self.upc5 = nn.Sequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
+        self.upc6 = nn.Sequential(
+            vgg_layer(64 * 2, 64),
+            nn.ConvTranspose2d(64, nc, 3, 1, 1),
+            nn.Sigmoid(),
+        )
+        self.up = nn.UpsamplingNearest2d(scale_factor=2)
+
+    def forward(self, inp):
+        vec, skip = inp
+        d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
+        d2 = self.upc2(torch.cat([self.up(d1), skip[4]], 1))
+        d3 = self.upc3(torch.cat([self.up(d2), skip[3]], 1))
+        d4 = self.upc4(torch.cat([self.up(d3), skip[2]], 1))
+        d5 = self.upc5(torch.cat([self.up(d4), skip[1]], 1))
+        return self.upc6(torch.cat([self.up(d5), skip[0]], 1))

@@ -1,0 +1,103 @@
+"""Recurrent stacks: `lstm` and `gaussian_lstm`.
+
+Same capability as the reference stacks (reference models/lstm.py:5-94):
+embed Linear -> N x LSTMCell -> output head (Linear+Tanh, or mu/logvar heads with
+reparameterization). Differences by design:
+
+- Device-agnostic: hidden state is created on the input's device/dtype at
+  `init_hidden` time. The reference hard-codes `.cuda()` at construction
+  (reference models/lstm.py:24-25,58,63-64), which makes CPU runs impossible;
+  we do not.
+- The per-timestep cell math dispatches to the fused HIP cell kernel on gfx950
+  (p2pvg_amd.ops.lstm_cell) and to torch.nn.functional on CPU.
+- Hidden state is module-owned and re-initialized per sequence, exactly like the
+  reference.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+
+class LSTMStack(nn.Module):
+    """Shared base: embed -> n_layers x LSTMCell."""
+
+    def __init__(self, input_size: int, hidden_size: int, n_layers: int):
+        super().__init__()
+        self.input_size = input_size
+        self.hidden_size = hidden_size
+        self.n_layers = n_layers
+        self.embed = nn.Linear(input_size, hidden_size)
+        self.lstm = nn.ModuleList(
+            [nn.LSTMCell(hidden_size, hidden_size) for _ in range(n_layers)]
+        )
+        self.hidden: Optional[List[Tuple[torch.Tensor, torch.Tensor]]] = None
+
+    def init_hidden(self, batch_size: int = 1, device=None, dtype=None):
+        p = next(self.parameters())
+        device = device if device is not None else p.device
+        dtype = dtype if dtype is not None else p.dtype
+        self.hidden = [
+            (
+                torch.zeros(batch_size, self.hidden_size, device=device, dtype=dtype),
+                torch.zeros(batch_size, self.hidden_size, device=device, dtype=dtype),
+            )
+            for _ in range(self.n_layers)
+        ]
+        return self.hidden
+
+    def _run_cells(self, x: torch.Tensor) -> torch.Tensor:
+        assert self.hidden is not None, "call init_hidden() before forward()"
+        h_in = self.embed(x.view(-1, self.input_size))
+        for i, cell in enumerate(self.lstm):
+            self.hidden[i] = ops.lstm_cell(
+                h_in,
+                self.hidden[i],
+                cell.weight_ih,
+                cell.weight_hh,
+                cell.bias_ih,
+                cell.bias_hh,
+            )
+            h_in = self.hidden[i][0]
+        return h_in
+
+
+class lstm(LSTMStack):
+    """Deterministic stack with Linear+Tanh output head (reference models/lstm.py:5-44)."""
+
+    def __init__(self, input_size, output_size, hidden_size, n_layers, batch_size=None):
+        super().__init__(input_size, hidden_size, n_layers)
+        self.output_size = output_size
+        self.batch_size = batch_size  # kept for checkpoint/API parity; unused
+        self.output = nn.Sequential(nn.Linear(hidden_size, output_size), nn.Tanh())
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.output(self._run_cells(x))
+
+
+class gaussian_lstm(LSTMStack):
+    """Stochastic stack: mu/logvar heads + reparameterization
+    (reference models/lstm.py:46-94)."""
+
+    def __init__(self, input_size, output_size, hidden_size, n_layers, batch_size=None):
+        super().__init__(input_size, hidden_size, n_layers)
+        self.output_size = output_size
+        self.batch_size = batch_size
+        self.mu_net = nn.Linear(hidden_size, output_size)
+        self.logvar_net = nn.Linear(hidden_size, output_size)
+
+    def reparameterize(self, mu: torch.Tensor, logvar: torch.Tensor) -> torch.Tensor:
+        std = logvar.mul(0.5).exp()
+        eps = torch.randn_like(std)
+        return eps * std + mu
+
+    def forward(self, x: torch.Tensor):
+        h = self._run_cells(x)
+        mu = self.mu_net(h)
+        logvar = self.logvar_net(h)
+        z = self.reparameterize(mu, logvar)
+        return z, mu, logvar

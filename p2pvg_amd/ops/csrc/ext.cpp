@@ -1,0 +1,33 @@
+// Python bindings for the p2pvg_amd gfx950 kernel library.
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<torch::Tensor> lstm_cell_fwd(torch::Tensor x, torch::Tensor h,
+                                         torch::Tensor c, torch::Tensor w_ih,
+                                         torch::Tensor w_hh, torch::Tensor b_ih,
+                                         torch::Tensor b_hh);
+std::vector<torch::Tensor> lstm_cell_bwd_pointwise(
+    torch::Tensor dh, c10::optional<torch::Tensor> dc_in, torch::Tensor gates,
+    torch::Tensor c_prev, torch::Tensor c_new);
+void multi_tensor_adam(std::vector<torch::Tensor> params,
+                       std::vector<torch::Tensor> grads,
+                       std::vector<torch::Tensor> exp_avgs,
+                       std::vector<torch::Tensor> exp_avg_sqs, double lr,
+                       double beta1, double beta2, double eps,
+                       double weight_decay, long step);
+torch::Tensor gaussian_kl_fwd(torch::Tensor mu1, torch::Tensor lv1,
+                              torch::Tensor mu2, torch::Tensor lv2,
+                              double denom);
+std::vector<torch::Tensor> gaussian_kl_bwd(torch::Tensor mu1, torch::Tensor lv1,
+                                           torch::Tensor mu2, torch::Tensor lv2,
+                                           torch::Tensor dout, double denom);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (gfx950)");
+  m.def("lstm_cell_bwd_pointwise", &lstm_cell_bwd_pointwise,
+        "LSTM cell backward pointwise (gfx950)");
+  m.def("multi_tensor_adam", &multi_tensor_adam, "multi-tensor Adam (gfx950)");
+  m.def("gaussian_kl_fwd", &gaussian_kl_fwd, "fused gaussian KL fwd (gfx950)");
+  m.def("gaussian_kl_bwd", &gaussian_kl_bwd, "fused gaussian KL bwd (gfx950)");
+}

@@ -1,0 +1,113 @@
+"""Model-level CPU tests: training step semantics, losses, generation."""
+import numpy as np
+import pytest
+import torch
+
+from p2pvg_amd.core import Config
+from p2pvg_amd.models import P2PModel
+
+
+def make_batch(cfg, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    return torch.rand(
+        cfg.max_seq_len, cfg.batch_size, cfg.channels, cfg.image_width, cfg.image_width,
+        generator=g,
+    )
+
+
+def test_forward_returns_four_losses(tiny_cfg):
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(tiny_cfg)
+    x = make_batch(tiny_cfg)
+    mse, kld, cpc, align = model(x, 0, len(x) - 1)
+    for v in (mse, kld, cpc, align):
+        assert torch.is_tensor(v) and v.dim() == 0
+        assert torch.isfinite(v)
+    assert mse > 0
+    assert kld > 0
+    assert cpc > 0  # cp step always processed
+
+
+def test_training_decreases_loss(tiny_cfg):
+    torch.manual_seed(0)
+    np.random.seed(0)
+    tiny_cfg.skip_prob = 0.0
+    tiny_cfg.lr = 1e-3
+    model = P2PModel(tiny_cfg)
+    x = make_batch(tiny_cfg)
+    first = None
+    last = None
+    for i in range(8):
+        model.zero_grad(set_to_none=False)
+        mse, _, _, _ = model(x, 0, len(x) - 1)
+        if first is None:
+            first = mse.item()
+        last = mse.item()
+    assert last < first, f"mse did not decrease: {first} -> {last}"
+
+
+def test_skip_gate_semantics(tiny_cfg):
+    """Skip gate must never skip i==1 or the cp step, and bounds skips."""
+    torch.manual_seed(0)
+    tiny_cfg.skip_prob = 1.0  # always want to skip where legal
+    model = P2PModel(tiny_cfg)
+    x = make_batch(tiny_cfg)
+    np.random.seed(0)
+    mse, kld, cpc, align = model(x, 0, len(x) - 1)
+    # cpc computed means cp step ran
+    assert cpc > 0
+
+
+def test_generate_modes_and_lengths(tiny_cfg):
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(tiny_cfg)
+    model.eval()
+    x = make_batch(tiny_cfg)
+    for mode in ("full", "posterior", "prior"):
+        for length in (5, 8, 12):
+            out = model.p2p_generate(x, length, length - 1, model_mode=mode)
+            assert len(out) == length
+            for f in out:
+                assert f.shape == x[0].shape
+
+
+def test_generate_skip_frame_inserts_zeros(tiny_cfg):
+    torch.manual_seed(0)
+    model = P2PModel(tiny_cfg)
+    model.eval()
+    tiny_cfg.skip_prob = 1.0
+    x = make_batch(tiny_cfg)
+    np.random.seed(3)
+    out = model.p2p_generate(x, 12, 11, model_mode="full", skip_frame=True)
+    assert len(out) == 12
+    zeros = sum(1 for f in out if f.abs().sum() == 0)
+    assert zeros >= 1  # some frames skipped as placeholders
+
+
+def test_mlp_backbone_h36m_path():
+    cfg = Config(
+        dataset="h36m", backbone="mlp", batch_size=2, max_seq_len=8, delta_len=1,
+        g_dim=32, z_dim=4, rnn_size=32, device="cpu",
+    )
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(cfg)
+    pose_3d = torch.randn(8, 2, 17, 3)
+    pose_2d = pose_3d[..., :2].clone()
+    losses = model((pose_2d, pose_3d, [0, 1]), 0, 7)
+    assert all(torch.isfinite(v) for v in losses)
+
+
+@pytest.mark.parametrize("backbone,width", [("dcgan", 64), ("dcgan", 128), ("vgg", 64), ("vgg", 128)])
+def test_all_backbones_shapes(backbone, width):
+    from p2pvg_amd.models.backbones import get_backbone
+
+    enc_cls, dec_cls = get_backbone(backbone, width, "bair")
+    enc, dec = enc_cls(32, 3), dec_cls(32, 3)
+    x = torch.randn(2, 3, width, width)
+    latent, skips = enc(x)
+    assert latent.shape == (2, 32)
+    out = dec([latent, skips])
+    assert out.shape == x.shape
