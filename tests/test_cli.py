@@ -1,0 +1,84 @@
+"""End-to-end CLI tests: train.py one tiny epoch on CPU, then generate.py
+consumes the written checkpoint (the §3.4 round-trip through both CLIs)."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.slow
+def test_train_then_generate(tmp_path):
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    log_dir = tmp_path / "logs" / "run"
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "train.py"),
+         "--dataset", "mnist", "--backbone", "dcgan", "--batch_size", "2",
+         "--max_seq_len", "6", "--delta_len", "1", "--g_dim", "32",
+         "--z_dim", "4", "--rnn_size", "32", "--nepochs", "1",
+         "--epoch_size", "3", "--nsample", "2", "--device", "cpu",
+         "--data_root", "/nonexistent", "--num_workers", "0",
+         "--log_dir", str(log_dir)],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-3000:]
+
+    # locate the run dir (train.py appends the hyperparam suffix)
+    runs = [d for d in (tmp_path / "logs").iterdir() if d.is_dir()]
+    assert runs, "no log dir created"
+    run = runs[0]
+    ckpt = run / "model.pth"
+    assert ckpt.exists(), f"missing latest checkpoint in {run}"
+    assert (run / "model_0.pth").exists()
+    assert (run / "cmd.txt").exists()
+    assert (run / "scalars.jsonl").exists()
+    vis = list((run / "gen_vis").glob("*.png"))
+    assert vis, "no qualitative eval images written"
+    gifs = list((run / "gen_vis").glob("*.gif"))
+    assert gifs, "no GIFs written"
+
+    # generate.py consumes the checkpoint with a frame-dir input
+    frames_dir = tmp_path / "frames"
+    frames_dir.mkdir()
+    import numpy as np
+    from PIL import Image
+
+    for i in range(6):
+        arr = (np.random.rand(64, 64, 3) * 255).astype("uint8")
+        Image.fromarray(arr).save(frames_dir / f"{i:02d}.png")
+
+    out_dir = tmp_path / "gen_out"
+    r2 = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "generate.py"),
+         "--ckpt", str(ckpt), "--video", str(frames_dir),
+         "--output_root", str(out_dir), "--device", "cpu"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r2.returncode == 0, r2.stderr[-3000:]
+    for length in (10, 20, 30):
+        assert (out_dir / f"len_{length}-gt.png").exists()
+        assert (out_dir / f"len_{length}-gen_full.gif").exists()
+
+
+@pytest.mark.slow
+def test_bench_contract_cpu(tmp_path):
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "bench.py"),
+         "--steps", "2", "--warmup", "1", "--batch", "2", "--seq_len", "6"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-3000:]
+    line = r.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, f"bench JSON missing {key}"
+    assert d["data"] == "synthetic"
+    assert d["config"]["global_batch"] == 2
