@@ -50,6 +50,9 @@ def main():
     p.add_argument("--dataset", type=str, default="bair")
     p.add_argument("--dtype", type=str, default="bf16")
     p.add_argument("--kernels", type=str, default="auto")
+    p.add_argument("--channels_last", type=int, default=1,
+                   help="NHWC activations/weights (MIOpen igemm is NHWC-native)")
+    p.add_argument("--use_graphs", type=int, default=0)
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -86,6 +89,7 @@ def main():
         dtype=args.dtype if use_cuda else "fp32",
         device=str(device),
         ddp=world > 1,
+        use_graphs=bool(args.use_graphs) and use_cuda,
     )
     if args.dataset == "h36m":
         cfg.backbone = "mlp"
@@ -94,6 +98,8 @@ def main():
     np.random.seed(1234 + rank)
 
     model = P2PModel(cfg).to(device)
+    if args.channels_last and use_cuda and args.dataset != "h36m":
+        model = model.to(memory_format=torch.channels_last)
     if world > 1:
         from p2pvg_amd.parallel import DDPGradSync
 
@@ -101,15 +107,33 @@ def main():
         sync.broadcast_parameters()
 
     x = make_synthetic_batch(cfg, device, seed=1234 + rank)
+    if args.channels_last and use_cuda and args.dataset != "h36m":
+        # store the clip NHWC so each x[i] is a channels_last (B,C,H,W) view
+        t, b, c, h, w = x.shape
+        nhwc = torch.empty(t, b, h, w, c, device=device, dtype=x.dtype)
+        nhwc.copy_(x.permute(0, 1, 3, 4, 2))
+        x = nhwc.permute(0, 1, 4, 2, 3)
     amp = cfg.dtype == "bf16" and use_cuda
 
-    def one_step():
-        model.zero_grad(set_to_none=False)
-        if amp:
-            with torch.autocast("cuda", dtype=torch.bfloat16):
+    if cfg.use_graphs:
+        from p2pvg_amd.runtime import GraphedTrainStep
+
+        stepper = GraphedTrainStep(
+            model, amp_dtype=torch.bfloat16 if amp else None
+        )
+
+        def one_step():
+            stepper.step(x)
+
+    else:
+
+        def one_step():
+            model.zero_grad(set_to_none=False)
+            if amp:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    model(x, 0, len(x) - 1)
+            else:
                 model(x, 0, len(x) - 1)
-        else:
-            model(x, 0, len(x) - 1)
 
     # warmup
     for _ in range(args.warmup):

@@ -33,6 +33,34 @@ from .backbones import build_backbone
 from .lstm import gaussian_lstm, lstm
 
 
+from dataclasses import dataclass
+
+
+@dataclass(frozen=True)
+class StepPlan:
+    """Host-side plan of one training step (which steps run, time signals)."""
+
+    seq_len: int
+    cp_ix: int
+    proc: tuple          # processed step indices i (i=1..cp_ix, minus skips)
+    tun: np.ndarray      # time_until_cp per processed step
+    dts: np.ndarray      # delta_time per processed step
+    unpack: tuple        # per processed step: does this step refresh `skip`?
+
+    @property
+    def graph_key(self):
+        return (self.seq_len, len(self.proc), self.unpack)
+
+
+def gather_frames(x: torch.Tensor, idx: torch.Tensor) -> torch.Tensor:
+    """index_select frames along time, preserving per-frame channels_last."""
+    if x.dim() == 5 and x[0].is_contiguous(memory_format=torch.channels_last):
+        xn = x.permute(0, 1, 3, 4, 2)          # (T,B,H,W,C) view of NHWC storage
+        sel = xn.index_select(0, idx)          # contiguous (n,B,H,W,C)
+        return sel.permute(0, 1, 4, 2, 3)      # (n,B,C,H,W), channels_last frames
+    return x.index_select(0, idx)
+
+
 def init_weights(m: nn.Module) -> None:
     """N(0, 0.02) conv/linear, N(1, 0.02) batchnorm (reference misc/utils.py:157-164)."""
     classname = m.__class__.__name__
@@ -81,9 +109,12 @@ class P2PModel(nn.Module):
         cfg = self.cfg
         from ..optim import make_adam
 
+        device = cfg.resolved_device() if hasattr(cfg, "resolved_device") else "cpu"
         for name in ("frame_predictor", "posterior", "prior", "encoder", "decoder"):
             opt = make_adam(
-                getattr(self, name).parameters(), lr=cfg.lr, betas=(cfg.beta1, 0.999)
+                getattr(self, name).parameters(), lr=cfg.lr,
+                betas=(cfg.beta1, 0.999), device=device,
+                capturable=bool(getattr(cfg, "use_graphs", False)),
             )
             setattr(self, f"{name}_optimizer", opt)
 
@@ -105,10 +136,127 @@ class P2PModel(nn.Module):
         h_cp = self.encoder(x_cp)[0]
         return x_cp, h_cp
 
-    def _skip_gate(self, seq_len: int, cp_ix: int, length: int) -> np.ndarray:
-        """Pre-draw the per-step skip probabilities on host (kept on host for
-        parity with reference models/p2p_model.py:215)."""
-        return np.random.uniform(0, 1, length - 1)
+    def plan_step(self, seq_len: int) -> "StepPlan":
+        """Host-side step plan: draw the skip gate and derive the processed
+        step indices and the two time-signal values per processed step.
+
+        Skip-gate semantics match reference models/p2p_model.py:209-222: one
+        pre-drawn U(0,1) per candidate step; never skips i==1 or the
+        control-point step; at most seq_len*skip_prob skips; skipped steps
+        freeze hidden state and widen the next delta_time.
+        """
+        cfg = self.cfg
+        cp_ix = seq_len - 1
+        probs = np.random.uniform(0, 1, seq_len - 1)
+        proc, tun, dts = [], [], []
+        prev_i = 0
+        skip_count = 0
+        max_skip = seq_len * cfg.skip_prob
+        for i in range(1, seq_len):
+            if (
+                probs[i - 1] <= cfg.skip_prob
+                and i >= cfg.n_past
+                and skip_count < max_skip
+                and i != 1
+                and i != cp_ix
+            ):
+                skip_count += 1
+                continue
+            proc.append(i)
+            tun.append((cp_ix - i + 1) / cp_ix)
+            dts.append((i - prev_i) / cp_ix)
+            prev_i = i
+        unpack = tuple(
+            bool(cfg.last_frame_skip or i <= cfg.n_past) for i in proc
+        )
+        return StepPlan(
+            seq_len=seq_len,
+            cp_ix=cp_ix,
+            proc=tuple(proc),
+            tun=np.asarray(tun, dtype=np.float32),
+            dts=np.asarray(dts, dtype=np.float32),
+            unpack=unpack,
+        )
+
+    def _compute_losses(self, prev_frames, cur_frames, tun_t, dt_t, plan):
+        """Device-side training losses over the processed steps.
+
+        prev_frames/cur_frames: (n, B, ...) gathered frames (prev_frames[k] =
+        x[proc[k]-1], cur_frames[k] = x[proc[k]]); tun_t/dt_t: (n,1,1) device
+        scalars. The structure depends only on (n, plan.unpack) — it is
+        hipGraph-capturable with the frames/scalars as graph inputs.
+        """
+        cfg = self.cfg
+        n = len(plan.proc)
+        batch_size = prev_frames[0].shape[0]
+        device = prev_frames.device
+
+        self.init_hidden(batch_size=batch_size, device=device)
+
+        x_cp = cur_frames[n - 1]
+        global_z = self.encoder(x_cp)[0]
+
+        mse_loss = torch.zeros((), device=device)
+        kld_loss = torch.zeros((), device=device)
+        cpc_loss = torch.zeros((), device=device)
+        align_loss = torch.zeros((), device=device)
+
+        h_prev = None
+        h_pred = None
+        skip = None
+
+        for k in range(n):
+            if k > 0 and cfg.align_mode == "reference":
+                # exact as-written semantics: batch row 0 of the PREVIOUS
+                # iteration's latent broadcast against h_pred
+                # (reference models/p2p_model.py:225 quirk, SURVEY §2.2)
+                align_loss = align_loss + self.align_criterion(
+                    h_prev[0].expand_as(h_pred), h_pred
+                )
+
+            time_until_cp = tun_t[k].expand(batch_size, 1)
+            delta_time = dt_t[k].expand(batch_size, 1)
+
+            h = self.encoder(prev_frames[k])
+            h_target = self.encoder(cur_frames[k])[0]
+
+            if plan.unpack[k]:
+                h, skip = h
+            else:
+                h = h[0]
+
+            if k > 0 and cfg.align_mode == "paper":
+                # paper-intent: align the encoder latent of frame i-1 with
+                # the predictor output that predicted that frame's latent
+                align_loss = align_loss + self.align_criterion(h, h_pred)
+
+            h_cpaw = torch.cat([h, global_z, time_until_cp, delta_time], 1)
+            h_target_cpaw = torch.cat(
+                [h_target, global_z, time_until_cp, delta_time], 1
+            )
+
+            zt, mu, logvar = self.posterior(h_target_cpaw)
+            zt_p, mu_p, logvar_p = self.prior(h_cpaw)
+
+            h_pred = self.frame_predictor(
+                torch.cat([h, zt, time_until_cp, delta_time], 1)
+            )
+            x_pred = self.decoder([h_pred, skip])
+
+            if k == n - 1:
+                # control-point consistency (reference models/p2p_model.py:251-254)
+                h_pred_p = self.frame_predictor(
+                    torch.cat([h, zt_p, time_until_cp, delta_time], 1)
+                )
+                x_pred_p = self.decoder([h_pred_p, skip])
+                cpc_loss = self.mse_criterion(x_pred_p, x_cp)
+
+            mse_loss = mse_loss + self.mse_criterion(x_pred, cur_frames[k])
+            kld_loss = kld_loss + self.kl_criterion(mu, logvar, mu_p, logvar_p)
+
+            h_prev = h
+
+        return mse_loss, kld_loss, cpc_loss, align_loss
 
     # -- training step ------------------------------------------------------
 
@@ -119,109 +267,29 @@ class P2PModel(nn.Module):
         if isinstance(x, tuple):  # h36m yields (pose_2d, pose_3d, camera_view)
             x = x[1]
 
-        cfg = self.cfg
-        batch_size = x[0].shape[0]
         device = x[0].device
-
-        self.init_hidden(batch_size=batch_size, device=device)
-
         seq_len = len(x)
-        cp_ix = seq_len - 1
-        x_cp, global_z = self.get_global_descriptor(x, 0, cp_ix)
+        plan = self.plan_step(seq_len)
 
-        mse_loss = torch.zeros((), device=device)
-        kld_loss = torch.zeros((), device=device)
-        cpc_loss = torch.zeros((), device=device)
-        align_loss = torch.zeros((), device=device)
+        idx_cur = torch.tensor(plan.proc, device=device)
+        prev_frames = gather_frames(x, idx_cur - 1)
+        cur_frames = gather_frames(x, idx_cur)
+        tun_t = torch.as_tensor(plan.tun).to(device).view(-1, 1, 1)
+        dt_t = torch.as_tensor(plan.dts).to(device).view(-1, 1, 1)
 
-        skip_prob = cfg.skip_prob
-        max_skip_count = seq_len * skip_prob
-        skip_count = 0
-        prev_i = 0
-        probs = self._skip_gate(seq_len, cp_ix, seq_len)
-
-        h_prev = None       # previous iteration's (possibly stale) encoder latent
-        h_pred = None       # previous iteration's predictor output
-
-        for i in range(1, seq_len):
-            # skip gate: never skips i==1 or the control-point step
-            # (reference models/p2p_model.py:220)
-            if (
-                probs[i - 1] <= skip_prob
-                and i >= cfg.n_past
-                and skip_count < max_skip_count
-                and i != 1
-                and i != cp_ix
-            ):
-                skip_count += 1
-                continue
-
-            # latent alignment loss (see SURVEY §2.2 for the as-written quirk)
-            if i > 1 and h_pred is not None:
-                if cfg.align_mode == "reference":
-                    # exact as-written semantics: batch row 0 of the PREVIOUS
-                    # iteration's latent broadcast against h_pred
-                    # (reference models/p2p_model.py:225)
-                    align_loss = align_loss + self.align_criterion(
-                        h_prev[0].expand_as(h_pred), h_pred
-                    )
-                # paper-mode align is added after encoding x[i-1] below
-
-            time_until_cp = torch.full(
-                (batch_size, 1), (cp_ix - i + 1) / cp_ix, device=device, dtype=x_cp.dtype
-            )
-            delta_time = torch.full(
-                (batch_size, 1), (i - prev_i) / cp_ix, device=device, dtype=x_cp.dtype
-            )
-            prev_i = i
-
-            h = self.encoder(x[i - 1])
-            h_target = self.encoder(x[i])[0]
-
-            if cfg.last_frame_skip or i <= cfg.n_past:
-                h, skip = h
-            else:
-                h = h[0]
-
-            if i > 1 and h_pred is not None and cfg.align_mode == "paper":
-                # paper-intent: align the encoder latent of frame i-1 with the
-                # predictor output that predicted that same frame's latent
-                align_loss = align_loss + self.align_criterion(h, h_pred)
-
-            h_cpaw = torch.cat([h, global_z, time_until_cp, delta_time], 1)
-            h_target_cpaw = torch.cat([h_target, global_z, time_until_cp, delta_time], 1)
-
-            zt, mu, logvar = self.posterior(h_target_cpaw)
-            zt_p, mu_p, logvar_p = self.prior(h_cpaw)
-
-            h_pred = self.frame_predictor(
-                torch.cat([h, zt, time_until_cp, delta_time], 1)
-            )
-            x_pred = self.decoder([h_pred, skip])
-
-            if i == cp_ix:
-                # control-point consistency: the prior-path generation of the
-                # end frame must match it (reference models/p2p_model.py:251-254)
-                h_pred_p = self.frame_predictor(
-                    torch.cat([h, zt_p, time_until_cp, delta_time], 1)
-                )
-                x_pred_p = self.decoder([h_pred_p, skip])
-                cpc_loss = self.mse_criterion(x_pred_p, x_cp)
-
-            mse_loss = mse_loss + self.mse_criterion(x_pred, x[i])
-            kld_loss = kld_loss + self.kl_criterion(mu, logvar, mu_p, logvar_p)
-
-            h_prev = h
-
+        mse_loss, kld_loss, cpc_loss, align_loss = self._compute_losses(
+            prev_frames, cur_frames, tun_t, dt_t, plan
+        )
         self._backward_and_step(mse_loss, kld_loss, cpc_loss, align_loss)
 
         inv = 1.0 / seq_len
         return (
             (mse_loss * inv).detach(),
             (kld_loss * inv).detach(),
-            (cpc_loss * inv).detach() if torch.is_tensor(cpc_loss) else cpc_loss,
+            (cpc_loss * inv).detach(),
             (align_loss * inv).detach(),
         )
+
 
     def _param_groups(self):
         nonprior = [p for n in self.NONPRIOR for p in getattr(self, n).parameters()]

@@ -6,6 +6,10 @@
 // exp_avg, exp_avg_sq pointers ride in the kernarg block; blocks grid-stride
 // within (tensor, chunk) space. fp32, bias-corrected, matching
 // torch.optim.Adam exactly.
+//
+// The step count is read from DEVICE memory (one fp32 scalar, incremented by
+// the caller with a tensor op) so the whole optimizer step is
+// hipGraph-capturable: no host-side state is baked into the launch.
 
 #include "common.h"
 
@@ -23,11 +27,9 @@ struct AdamArgs {
   long n[CHUNK_TENSORS];
 };
 
-__global__ __launch_bounds__(BLOCK) void adam_kernel(AdamArgs args, int ntensors,
-                                                     float lr, float beta1,
-                                                     float beta2, float eps,
-                                                     float weight_decay,
-                                                     float bc1, float bc2) {
+__global__ __launch_bounds__(BLOCK) void adam_kernel(
+    AdamArgs args, int ntensors, float lr, float beta1, float beta2, float eps,
+    float weight_decay, const float* __restrict__ step_ptr) {
   const int t = blockIdx.y;
   if (t >= ntensors) return;
   float* __restrict__ p = args.p[t];
@@ -36,6 +38,9 @@ __global__ __launch_bounds__(BLOCK) void adam_kernel(AdamArgs args, int ntensors
   float* __restrict__ v = args.v[t];
   const long n = args.n[t];
 
+  const float step = step_ptr[0];
+  const float bc1 = 1.f - __powf(beta1, step);
+  const float bc2 = 1.f - __powf(beta2, step);
   const float step_size = lr / bc1;
   const float inv_sqrt_bc2 = rsqrtf(bc2);
 
@@ -65,12 +70,13 @@ void multi_tensor_adam(std::vector<torch::Tensor> params,
                        std::vector<torch::Tensor> exp_avgs,
                        std::vector<torch::Tensor> exp_avg_sqs, double lr,
                        double beta1, double beta2, double eps,
-                       double weight_decay, long step) {
+                       double weight_decay, torch::Tensor step) {
   TORCH_CHECK(params.size() == grads.size() && params.size() == exp_avgs.size() &&
                   params.size() == exp_avg_sqs.size(),
               "multi_tensor_adam: list length mismatch");
-  const float bc1 = 1.f - powf((float)beta1, (float)step);
-  const float bc2 = 1.f - powf((float)beta2, (float)step);
+  CHECK_CUDA(step);
+  TORCH_CHECK(step.scalar_type() == torch::kFloat32,
+              "multi_tensor_adam: step must be a device fp32 scalar");
   auto stream = at::cuda::getCurrentCUDAStream();
 
   size_t i = 0;
@@ -80,9 +86,9 @@ void multi_tensor_adam(std::vector<torch::Tensor> params,
     long max_n = 0;
     for (; nt < CHUNK_TENSORS && i < params.size(); ++nt, ++i) {
       CHECK_INPUT(params[i]);
+      CHECK_INPUT(grads[i]);
       TORCH_CHECK(params[i].scalar_type() == torch::kFloat32,
                   "multi_tensor_adam: fp32 only");
-      CHECK_INPUT(grads[i]);
       args.p[nt] = params[i].data_ptr<float>();
       args.g[nt] = grads[i].data_ptr<float>();
       args.m[nt] = exp_avgs[i].data_ptr<float>();
@@ -90,10 +96,10 @@ void multi_tensor_adam(std::vector<torch::Tensor> params,
       args.n[nt] = params[i].numel();
       max_n = std::max(max_n, args.n[nt]);
     }
-    const int bx = std::min<long>(1024, (max_n + BLOCK * ILP - 1) / (BLOCK * ILP));
+    const int bx = std::min<long>(256, (max_n + BLOCK * ILP - 1) / (BLOCK * ILP));
     dim3 grid(std::max(bx, 1), nt);
     hipLaunchKernelGGL(adam_kernel, grid, dim3(BLOCK), 0, stream, args, nt,
                        (float)lr, (float)beta1, (float)beta2, (float)eps,
-                       (float)weight_decay, bc1, bc2);
+                       (float)weight_decay, step.data_ptr<float>());
   }
 }

@@ -15,7 +15,7 @@ void multi_tensor_adam(std::vector<torch::Tensor> params,
                        std::vector<torch::Tensor> exp_avgs,
                        std::vector<torch::Tensor> exp_avg_sqs, double lr,
                        double beta1, double beta2, double eps,
-                       double weight_decay, long step);
+                       double weight_decay, torch::Tensor step);
 torch::Tensor gaussian_kl_fwd(torch::Tensor mu1, torch::Tensor lv1,
                               torch::Tensor mu2, torch::Tensor lv2,
                               double denom);

@@ -1,0 +1,124 @@
+"""hipGraph capture of the whole training step (SURVEY §5.7 MI355X plan).
+
+The per-timestep recurrence launches thousands of small kernels per training
+step; at the measured batch sizes the GPU sits idle well over half the wall
+time between launches (profiles/ shows the kernel-trace evidence). Capturing
+the entire step — forward over all processed timesteps, both backward phases,
+and all five (capturable) Adam steps — into one hipGraph and replaying it
+removes every launch gap.
+
+Dynamic control flow is handled exactly as SURVEY §5.7 prescribes:
+- The skip gate and dynamic sequence length are HOST-side (plan_step); the
+  captured compute depends only on the plan's graph_key (seq_len, n_proc,
+  unpack pattern). One graph is captured per distinct key, on demand.
+- Which frames feed each step, and the two time-signal scalars, are GRAPH
+  INPUTS: the gathered (n,B,C,H,W) frame buffers and (n,1,1) scalar buffers
+  are static tensors refilled before each replay.
+- Reparameterization randomness uses the device philox generator, which
+  torch.cuda.CUDAGraph captures/advances correctly.
+
+Capture warmup runs the step function a few times on a side stream (the
+standard recipe: materializes grads, Adam state, MIOpen solutions, autocast
+weight casts) — those warmup iterations ARE real optimizer steps; for
+benchmarking they land in the warmup phase, for training they are ordinary
+extra steps on the first batch of that shape.
+"""
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+
+from ..models.p2p import StepPlan, gather_frames
+
+
+class _GraphEntry:
+    __slots__ = ("graph", "prev_buf", "cur_buf", "tun", "dts", "loss_out")
+
+
+class GraphedTrainStep:
+    def __init__(self, model, amp_dtype: Optional[torch.dtype] = None,
+                 warmup_iters: int = 3):
+        self.model = model
+        self.amp_dtype = amp_dtype
+        self.warmup_iters = warmup_iters
+        self.graphs: Dict[Tuple, _GraphEntry] = {}
+        self.pool = None  # shared memory pool across all captured graphs
+
+    def _inner(self, entry: _GraphEntry, plan: StepPlan):
+        model = self.model
+        model.zero_grad(set_to_none=False)
+        if self.amp_dtype is not None:
+            ctx = torch.autocast("cuda", dtype=self.amp_dtype, cache_enabled=False)
+        else:
+            import contextlib
+
+            ctx = contextlib.nullcontext()
+        with ctx:
+            losses = model._compute_losses(
+                entry.prev_buf, entry.cur_buf, entry.tun, entry.dts, plan
+            )
+        model._backward_and_step(*losses)
+        mse, kld, cpc, align = losses
+        entry.loss_out.copy_(
+            torch.stack([mse.detach(), kld.detach(), cpc.detach(), align.detach()])
+        )
+
+    def _capture(self, plan: StepPlan, prev: torch.Tensor, cur: torch.Tensor,
+                 tun: torch.Tensor, dts: torch.Tensor) -> _GraphEntry:
+        entry = _GraphEntry()
+        entry.prev_buf = prev.clone()
+        entry.cur_buf = cur.clone()
+        entry.tun = tun.clone()
+        entry.dts = dts.clone()
+        entry.loss_out = torch.zeros(4, device=prev.device)
+
+        # warmup on a side stream (standard CUDAGraph recipe)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(self.warmup_iters):
+                self._inner(entry, plan)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
+        g = torch.cuda.CUDAGraph()
+        if self.pool is None:
+            with torch.cuda.graph(g):
+                self._inner(entry, plan)
+            self.pool = g.pool()
+        else:
+            with torch.cuda.graph(g, pool=self.pool):
+                self._inner(entry, plan)
+        entry.graph = g
+        return entry
+
+    def step(self, x: torch.Tensor):
+        """One training step on x ((T,B,...) device tensor or h36m tuple)."""
+        model = self.model
+        if isinstance(x, tuple):
+            x = x[1]
+        seq_len = len(x)
+        plan = model.plan_step(seq_len)
+        device = x.device
+
+        idx_cur = torch.tensor(plan.proc, device=device)
+        prev = gather_frames(x, idx_cur - 1)
+        cur = gather_frames(x, idx_cur)
+        tun = torch.as_tensor(plan.tun).to(device).view(-1, 1, 1)
+        dts = torch.as_tensor(plan.dts).to(device).view(-1, 1, 1)
+
+        key = plan.graph_key
+        entry = self.graphs.get(key)
+        if entry is None:
+            # capture records but does not execute; fall through to replay
+            entry = self._capture(plan, prev, cur, tun, dts)
+            self.graphs[key] = entry
+
+        entry.prev_buf.copy_(prev)
+        entry.cur_buf.copy_(cur)
+        entry.tun.copy_(tun)
+        entry.dts.copy_(dts)
+        entry.graph.replay()
+        inv = 1.0 / seq_len
+        return tuple((entry.loss_out[i] * inv).clone() for i in range(4))

@@ -1,0 +1,62 @@
+"""hipGraph-captured training step: replay correctness on GPU."""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_graphed_step_trains():
+    from p2pvg_amd.core import Config
+    from p2pvg_amd.models import P2PModel
+    from p2pvg_amd.runtime import GraphedTrainStep
+
+    cfg = Config(dataset="bair", backbone="dcgan", channels=3, batch_size=2,
+                 max_seq_len=6, g_dim=32, z_dim=4, rnn_size=64, device="cuda",
+                 skip_prob=0.0, use_graphs=True)
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(cfg).to("cuda")
+    stepper = GraphedTrainStep(model, amp_dtype=None, warmup_iters=2)
+
+    x = torch.rand(6, 2, 3, 64, 64, device="cuda")
+    p0 = model.decoder.upc1[0].weight.detach().clone()
+
+    losses1 = stepper.step(x)
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(v) for v in losses1)
+    assert not torch.equal(p0, model.decoder.upc1[0].weight), "no update applied"
+
+    # second replay: different data must give different losses
+    x2 = torch.rand(6, 2, 3, 64, 64, device="cuda")
+    losses2 = stepper.step(x2)
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(v) for v in losses2)
+    assert float(losses1[0]) != float(losses2[0])
+
+    # repeated stepping decreases recon loss on a fixed batch
+    first = float(stepper.step(x)[0])
+    for _ in range(6):
+        last = float(stepper.step(x)[0])
+    assert last < first, f"graphed training not learning: {first} -> {last}"
+
+
+def test_graphed_step_dynamic_length_two_graphs():
+    from p2pvg_amd.core import Config
+    from p2pvg_amd.models import P2PModel
+    from p2pvg_amd.runtime import GraphedTrainStep
+
+    cfg = Config(dataset="mnist", backbone="dcgan", channels=1, batch_size=2,
+                 max_seq_len=8, g_dim=32, z_dim=4, rnn_size=64, device="cuda",
+                 skip_prob=0.0, use_graphs=True)
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(cfg).to("cuda")
+    stepper = GraphedTrainStep(model, amp_dtype=None, warmup_iters=2)
+
+    for T in (6, 8, 6, 8):
+        x = torch.rand(T, 2, 1, 64, 64, device="cuda")
+        losses = stepper.step(x)
+        torch.cuda.synchronize()
+        assert all(torch.isfinite(v) for v in losses)
+    assert len(stepper.graphs) == 2

@@ -126,13 +126,15 @@ def test_multi_tensor_adam_matches_torch(ext):
 
         exp_avgs = [opt_ref.state[p]["exp_avg"] for p in params_ref]
 
-    # hip path: maintain our own state
+    # hip path: maintain our own state (device-side step scalar)
     m = [torch.zeros_like(p) for p in params_hip]
     v = [torch.zeros_like(p) for p in params_hip]
+    step_t = torch.zeros((), device=dev)
     for step in range(1, 4):
+        step_t += 1
         ext.multi_tensor_adam(
             [p.data for p in params_hip], [g for g in grads], m, v,
-            1e-3, 0.9, 0.999, 1e-8, 0.0, step,
+            1e-3, 0.9, 0.999, 1e-8, 0.0, step_t,
         )
 
     for a, b in zip(params_ref, params_hip):
