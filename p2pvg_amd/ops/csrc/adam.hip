@@ -85,8 +85,16 @@ void multi_tensor_adam(std::vector<torch::Tensor> params,
     int nt = 0;
     long max_n = 0;
     for (; nt < CHUNK_TENSORS && i < params.size(); ++nt, ++i) {
-      CHECK_INPUT(params[i]);
-      CHECK_INPUT(grads[i]);
+      // elementwise update: any dense layout is fine (channels_last conv
+      // weights included) as long as param/grad/state share it
+      CHECK_CUDA(params[i]);
+      TORCH_CHECK(params[i].is_non_overlapping_and_dense(),
+                  "multi_tensor_adam: param must be dense");
+      TORCH_CHECK(grads[i].strides() == params[i].strides(),
+                  "multi_tensor_adam: grad layout must match param");
+      TORCH_CHECK(exp_avgs[i].strides() == params[i].strides() &&
+                      exp_avg_sqs[i].strides() == params[i].strides(),
+                  "multi_tensor_adam: state layout must match param");
       TORCH_CHECK(params[i].scalar_type() == torch::kFloat32,
                   "multi_tensor_adam: fp32 only");
       args.p[nt] = params[i].data_ptr<float>();

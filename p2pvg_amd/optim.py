@@ -57,7 +57,7 @@ class HIPFusedAdam(torch.optim.Optimizer):
                     state["exp_avg"] = torch.zeros_like(p)
                     state["exp_avg_sq"] = torch.zeros_like(p)
                 params.append(p)
-                grads.append(p.grad if p.grad.is_contiguous() else p.grad.contiguous())
+                grads.append(p.grad)
                 ms.append(state["exp_avg"])
                 vs.append(state["exp_avg_sq"])
                 steps.append(state["step"])
