@@ -23,7 +23,15 @@ std::vector<torch::Tensor> gaussian_kl_bwd(torch::Tensor mu1, torch::Tensor lv1,
                                            torch::Tensor mu2, torch::Tensor lv2,
                                            torch::Tensor dout, double denom);
 
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
+                              c10::optional<torch::Tensor> bias, long stride,
+                              long pad, long act);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd,
+        "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA)");
   m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (gfx950)");
   m.def("lstm_cell_bwd_pointwise", &lstm_cell_bwd_pointwise,
         "LSTM cell backward pointwise (gfx950)");

@@ -56,8 +56,15 @@ class HIPFusedAdam(torch.optim.Optimizer):
                     state["step"] = torch.zeros((), dtype=torch.float32, device=p.device)
                     state["exp_avg"] = torch.zeros_like(p)
                     state["exp_avg_sq"] = torch.zeros_like(p)
+                g = p.grad
+                if g.stride() != p.stride():
+                    # rare: autograd left a grad in a different dense layout
+                    # (e.g. a transposed or channels-mismatched view) — copy
+                    # into the param's layout so the fused kernel can walk
+                    # param/grad/state with one index
+                    g = torch.empty_like(p).copy_(g)
                 params.append(p)
-                grads.append(p.grad)
+                grads.append(g)
                 ms.append(state["exp_avg"])
                 vs.append(state["exp_avg_sq"])
                 steps.append(state["step"])
