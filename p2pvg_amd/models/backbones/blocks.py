@@ -12,6 +12,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ...ops.conv import Conv2d, ConvTranspose2d
+
 
 class dcgan_conv(nn.Module):
     """Conv2d(k4,s2,p1) + BatchNorm2d + LeakyReLU(0.2)."""
@@ -19,7 +21,7 @@ class dcgan_conv(nn.Module):
     def __init__(self, nin: int, nout: int):
         super().__init__()
         self.main = nn.Sequential(
-            nn.Conv2d(nin, nout, 4, 2, 1),
+            Conv2d(nin, nout, 4, 2, 1),
             nn.BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
         )
@@ -34,7 +36,7 @@ class dcgan_upconv(nn.Module):
     def __init__(self, nin: int, nout: int):
         super().__init__()
         self.main = nn.Sequential(
-            nn.ConvTranspose2d(nin, nout, 4, 2, 1),
+            ConvTranspose2d(nin, nout, 4, 2, 1),
             nn.BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
         )
@@ -49,7 +51,7 @@ class vgg_layer(nn.Module):
     def __init__(self, nin: int, nout: int):
         super().__init__()
         self.main = nn.Sequential(
-            nn.Conv2d(nin, nout, 3, 1, 1),
+            Conv2d(nin, nout, 3, 1, 1),
             nn.BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
         )

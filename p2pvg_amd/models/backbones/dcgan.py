@@ -10,6 +10,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ...ops.conv import Conv2d, ConvTranspose2d
+
 from .blocks import dcgan_conv, dcgan_upconv
 
 
@@ -23,7 +25,7 @@ class Encoder64(nn.Module):
         self.c3 = dcgan_conv(nf * 2, nf * 4)      # -> 256x8x8
         self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x4x4
         self.c5 = nn.Sequential(                  # -> dim x1x1
-            nn.Conv2d(nf * 8, dim, 4, 1, 0),
+            Conv2d(nf * 8, dim, 4, 1, 0),
             nn.BatchNorm2d(dim),
             nn.Tanh(),
         )
@@ -43,7 +45,7 @@ class Decoder64(nn.Module):
         self.dim = dim
         nf = 64
         self.upc1 = nn.Sequential(
-            nn.ConvTranspose2d(dim, nf * 8, 4, 1, 0),
+            ConvTranspose2d(dim, nf * 8, 4, 1, 0),
             nn.BatchNorm2d(nf * 8),
             nn.LeakyReLU(0.2, inplace=True),
         )
@@ -51,7 +53,7 @@ class Decoder64(nn.Module):
         self.upc3 = dcgan_upconv(nf * 4 * 2, nf * 2)
         self.upc4 = dcgan_upconv(nf * 2 * 2, nf)
         self.upc5 = nn.Sequential(
-            nn.ConvTranspose2d(nf * 2, nc, 4, 2, 1),
+            ConvTranspose2d(nf * 2, nc, 4, 2, 1),
             nn.Sigmoid(),
         )
 
@@ -77,7 +79,7 @@ class Encoder128(nn.Module):
         self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x8
         self.c5 = dcgan_conv(nf * 8, nf * 8)      # -> 512x4
         self.c6 = nn.Sequential(
-            nn.Conv2d(nf * 8, dim, 4, 1, 0),
+            Conv2d(nf * 8, dim, 4, 1, 0),
             nn.BatchNorm2d(dim),
             nn.Tanh(),
         )
@@ -98,7 +100,7 @@ class Decoder128(nn.Module):
         self.dim = dim
         nf = 64
         self.upc1 = nn.Sequential(
-            nn.ConvTranspose2d(dim, nf * 8, 4, 1, 0),
+            ConvTranspose2d(dim, nf * 8, 4, 1, 0),
             nn.BatchNorm2d(nf * 8),
             nn.LeakyReLU(0.2, inplace=True),
         )
@@ -107,7 +109,7 @@ class Decoder128(nn.Module):
         self.upc4 = dcgan_upconv(nf * 4 * 2, nf * 2)
         self.upc5 = dcgan_upconv(nf * 2 * 2, nf)
         self.upc6 = nn.Sequential(
-            nn.ConvTranspose2d(nf * 2, nc, 4, 2, 1),
+            ConvTranspose2d(nf * 2, nc, 4, 2, 1),
             nn.Sigmoid(),
         )
 

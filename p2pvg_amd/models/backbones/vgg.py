@@ -10,6 +10,8 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ...ops.conv import Conv2d, ConvTranspose2d
+
 from .blocks import vgg_layer
 
 
@@ -26,7 +28,7 @@ class Encoder64(nn.Module):
             vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
         self.c5 = nn.Sequential(
-            nn.Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
+            Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
         )
         self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
 
@@ -44,7 +46,7 @@ class Decoder64(nn.Module):
         super().__init__()
         self.dim = dim
         self.upc1 = nn.Sequential(
-            nn.ConvTranspose2d(dim, 512, 4, 1, 0),
+            ConvTranspose2d(dim, 512, 4, 1, 0),
             nn.BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
@@ -57,7 +59,7 @@ class Decoder64(nn.Module):
         self.upc4 = nn.Sequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
         self.upc5 = nn.Sequential(
             vgg_layer(64 * 2, 64),
-            nn.ConvTranspose2d(64, nc, 3, 1, 1),
+            ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
         )
         self.up = nn.UpsamplingNearest2d(scale_factor=2)
@@ -87,7 +89,7 @@ class Encoder128(nn.Module):
             vgg_layer(512, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
         self.c6 = nn.Sequential(
-            nn.Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
+            Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
         )
         self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
 
@@ -106,7 +108,7 @@ class Decoder128(nn.Module):
         super().__init__()
         self.dim = dim
         self.upc1 = nn.Sequential(
-            nn.ConvTranspose2d(dim, 512, 4, 1, 0),
+            ConvTranspose2d(dim, 512, 4, 1, 0),
             nn.BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
@@ -122,7 +124,7 @@ class Decoder128(nn.Module):
         self.upc5 = nn.Sequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
         self.upc6 = nn.Sequential(
             vgg_layer(64 * 2, 64),
-            nn.ConvTranspose2d(64, nc, 3, 1, 1),
+            ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
         )
         self.up = nn.UpsamplingNearest2d(scale_factor=2)

@@ -1,9 +1,14 @@
-// NHWC implicit-GEMM forward convolution on MFMA for gfx950.
+// NHWC implicit-GEMM convolution on MFMA for gfx950.
 //
-// Covers the conv workloads of SURVEY §2.6 K1 (dispatch sites
+// Covers the conv workloads of SURVEY §2.6 K1-K3 (dispatch sites
 // reference models/vgg_64.py:8 k3s1p1, models/dcgan_64.py:8 k4s2p1,
-// encoder tails k4s1p0): out[n,ho,wo,k] = act(bias[k] +
-//   sum_{r,s,c} in[n, ho*S-P+r, wo*S-P+s, c] * w[k,r,s,c]).
+// encoder tails k4s1p0, decoder ConvTranspose k4s2p1):
+//   out[n,yo,xo,k] = act(bias[k] + sum_{r,s,c} in[n, ho*S-P+r, wo*S-P+s, c]
+//                                              * w[k,r,s,c])
+// where (yo,xo) = (ho*OYS+OY0, wo*OXS+OX0): OYS=1 is plain convolution;
+// OYS=2 with a per-parity weight slice realizes fractionally-strided
+// convolution (ConvTranspose fwd / stride-2 dgrad) as 4 dense sub-problems —
+// the xGMI-free, transpose-free CDNA4 formulation of K2/K3.
 //
 // GEMM view: M = N*HO*WO output pixels, Ndim = K output channels,
 // Kdim = R*S*C, iterated as (r,s) outer x 64-wide c-chunks inner so the
@@ -44,22 +49,27 @@ __device__ __forceinline__ float activate(float v, int act) {
   }
 }
 
+struct ScatterSpec {
+  int OH, OW;    // full output spatial dims
+  int oys, oy0, ox0;  // y = ho*oys + oy0, x = wo*oys + ox0
+};
+
 // in:  (N, H, W, C) bf16   w: (K, R, S, C) bf16   bias: (K) f32 or null
-// out: (N, HO, WO, K) bf16
+// out: (N, OH, OW, K) bf16, written at the scattered (y,x) positions
 template <int KSIZE, int STRIDE>
 __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ w,
     const float* __restrict__ bias, __bf16* __restrict__ out,
-    int Nb, int H, int W, int C, int K, int HO, int WO, int PAD, int act,
-    int mblocks) {
-  // LDS: A tile (BM x BK) + B tile (BN x BK), single-buffered
-  __shared__ __align__(16) char lds[(BM + BN) * ROW_BYTES + BM * 12];
+    int Nb, int H, int W, int C, int K, int HO, int WO, int PADH, int PADW,
+    int act, int mblocks, ScatterSpec sc) {
+  // LDS: A tile (BM x BK) + B tile (BN x BK), single-buffered, + pixel meta
+  __shared__ __align__(16) char lds[(BM + BN) * ROW_BYTES + BM * 16];
   char* a_lds = lds;
   char* b_lds = lds + BM * ROW_BYTES;
-  // per-pixel meta: input base offset (int, elements), hi0, wi0 (short)
-  int* pix_off = reinterpret_cast<int*>(lds + (BM + BN) * ROW_BYTES);
+  long* pix_out = reinterpret_cast<long*>(lds + (BM + BN) * ROW_BYTES);
+  int* pix_off = reinterpret_cast<int*>(pix_out + BM);
   short* pix_hi = reinterpret_cast<short*>(pix_off + BM);
-  short* pix_wi = reinterpret_cast<short*>(pix_hi + BM);
+  short* pix_wi = pix_hi + BM;
 
   // XCD-aware remap of the M dimension (bijective variant, guide §5)
   int bm_lin = blockIdx.x;
@@ -83,12 +93,17 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
       const int ho = rem / WO;
       const int wo = rem - ho * WO;
       pix_off[i] = n * H * W * C;
-      pix_hi[i] = (short)(ho * STRIDE - PAD);
-      pix_wi[i] = (short)(wo * STRIDE - PAD);
+      pix_hi[i] = (short)(ho * STRIDE - PADH);
+      pix_wi[i] = (short)(wo * STRIDE - PADW);
+      pix_out[i] =
+          (((long)n * sc.OH + (ho * sc.oys + sc.oy0)) * sc.OW +
+           (wo * sc.oys + sc.ox0)) *
+          K;
     } else {
       pix_off[i] = 0;
       pix_hi[i] = (short)-30000;  // always out of bounds -> zero rows
       pix_wi[i] = (short)-30000;
+      pix_out[i] = -1;
     }
   }
   __syncthreads();
@@ -104,7 +119,6 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int RSC = KSIZE * KSIZE * C;
   const bool cvec = (C % 8) == 0;
 
   for (int r = 0; r < KSIZE; ++r) {
@@ -190,10 +204,11 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     for (int i = 0; i < 4; ++i) {
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
-        const int mrow = m0 + wm + i * 16 + (lane >> 4) * 4 + v;
-        if (mrow < M) {
+        const int lrow = wm + i * 16 + (lane >> 4) * 4 + v;
+        const long ooff = pix_out[lrow];
+        if (ooff >= 0) {
           const float val = activate(acc[i][j][v] + bv, act);
-          out[(long)mrow * K + col] = (__bf16)val;
+          out[ooff + col] = (__bf16)val;
         }
       }
     }
@@ -203,8 +218,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 template <int KSIZE, int STRIDE>
 void launch_fwd(const torch::Tensor& in, const torch::Tensor& w,
                 const c10::optional<torch::Tensor>& bias, torch::Tensor& out,
-                int Nb, int H, int W, int C, int K, int HO, int WO, int PAD,
-                int act) {
+                int Nb, int H, int W, int C, int K, int HO, int WO, int PADH,
+                int PADW, int act, ScatterSpec sc) {
   const int M = Nb * HO * WO;
   const int mblocks = ceil_div(M, BM);
   dim3 grid(mblocks, ceil_div(K, BN));
@@ -215,7 +230,33 @@ void launch_fwd(const torch::Tensor& in, const torch::Tensor& w,
                      reinterpret_cast<const __bf16*>(w.data_ptr()),
                      bias.has_value() ? bias->data_ptr<float>() : nullptr,
                      reinterpret_cast<__bf16*>(out.data_ptr()), Nb, H, W, C, K,
-                     HO, WO, PAD, act, mblocks);
+                     HO, WO, PADH, PADW, act, mblocks, sc);
+}
+
+void dispatch_fwd(const torch::Tensor& in, const torch::Tensor& w,
+                  const c10::optional<torch::Tensor>& bias, torch::Tensor& out,
+                  int Nb, int H, int W, int C, int K, int HO, int WO, int R,
+                  int stride, int PADH, int PADW, int act, ScatterSpec sc) {
+  if (R == 3 && stride == 1) {
+    launch_fwd<3, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
+  } else if (R == 4 && stride == 2) {
+    launch_fwd<4, 2>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
+  } else if (R == 4 && stride == 1) {
+    launch_fwd<4, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
+  } else if (R == 2 && stride == 1) {
+    launch_fwd<2, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
+  } else if (R == 1 && stride == 1) {
+    launch_fwd<1, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
+  } else {
+    TORCH_CHECK(false, "unsupported conv geometry: k=", R, " stride=", stride);
+  }
+}
+
+void check_nhwc_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == torch::kBFloat16,
+              name, " must be a bf16 GPU tensor");
+  TORCH_CHECK(t.is_contiguous(at::MemoryFormat::ChannelsLast),
+              name, " must be channels_last");
 }
 
 }  // namespace
@@ -225,15 +266,8 @@ void launch_fwd(const torch::Tensor& in, const torch::Tensor& w,
 torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
                               c10::optional<torch::Tensor> bias, long stride,
                               long pad, long act) {
-  CHECK_CUDA(in);
-  CHECK_CUDA(w);
-  TORCH_CHECK(in.scalar_type() == torch::kBFloat16 &&
-                  w.scalar_type() == torch::kBFloat16,
-              "conv2d_nhwc_fwd: bf16 only");
-  TORCH_CHECK(in.is_contiguous(at::MemoryFormat::ChannelsLast),
-              "conv2d_nhwc_fwd: input must be channels_last");
-  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
-              "conv2d_nhwc_fwd: weight must be channels_last");
+  check_nhwc_bf16(in, "in");
+  check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
   const int K = w.size(0), R = w.size(2), S = w.size(3);
   TORCH_CHECK(w.size(1) == C, "channel mismatch");
@@ -247,17 +281,34 @@ torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
     CHECK_INPUT(bias.value());
     TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be fp32");
   }
-
-  if (R == 3 && stride == 1) {
-    launch_fwd<3, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, (int)pad, (int)act);
-  } else if (R == 4 && stride == 2) {
-    launch_fwd<4, 2>(in, w, bias, out, Nb, H, W, C, K, HO, WO, (int)pad, (int)act);
-  } else if (R == 4 && stride == 1) {
-    launch_fwd<4, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, (int)pad, (int)act);
-  } else if (R == 1 && stride == 1) {
-    launch_fwd<1, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, (int)pad, (int)act);
-  } else {
-    TORCH_CHECK(false, "unsupported conv geometry: k=", R, " stride=", stride);
-  }
+  ScatterSpec sc{HO, WO, 1, 0, 0};
+  dispatch_fwd(in, w, bias, out, Nb, H, W, C, K, HO, WO, R, (int)stride,
+               (int)pad, (int)pad, (int)act, sc);
   return out;
+}
+
+// Scatter variant: computes a stride-1 conv of `in` with `w` on a compact
+// HOxWO grid and writes results at out[:, :, oy0::oys, ox0::oxs]. `out` is
+// preallocated by the caller (one call per parity). ipad is the implicit
+// input padding of the compact problem.
+void conv2d_nhwc_fwd_scatter(torch::Tensor in, torch::Tensor w,
+                             c10::optional<torch::Tensor> bias,
+                             torch::Tensor out, long ipad_h, long ipad_w,
+                             long oys, long oy0, long ox0, long act) {
+  check_nhwc_bf16(in, "in");
+  check_nhwc_bf16(w, "w");
+  check_nhwc_bf16(out, "out");
+  const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C, "channel mismatch");
+  TORCH_CHECK(R == S, "square kernels only");
+  const int OH = out.size(2), OW = out.size(3);
+  // compact grid dims from the scatter spec
+  const int HO = (OH - 1 - (int)oy0) / (int)oys + 1;
+  const int WO = (OW - 1 - (int)ox0) / (int)oys + 1;
+  TORCH_CHECK(out.size(0) == Nb && out.size(1) == K, "bad out shape");
+
+  ScatterSpec sc{OH, OW, (int)oys, (int)oy0, (int)ox0};
+  dispatch_fwd(in, w, bias, out, Nb, H, W, C, K, HO, WO, R, 1, (int)ipad_h,
+               (int)ipad_w, (int)act, sc);
 }

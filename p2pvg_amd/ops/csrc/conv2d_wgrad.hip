@@ -1,0 +1,212 @@
+// NHWC weight-gradient (wgrad) kernel on MFMA for gfx950.
+//
+// Generic form covering conv wgrad AND ConvTranspose wgrad (roles swapped):
+//   dW[b, r, s, a] = sum_{n,ho,wo} Y[n,ho,wo,b] * X[n, ho*ST-P+r, wo*ST-P+s, a]
+// For conv: Y = grad_out, X = input, dW is the (K,R,S,C)-physical weight grad.
+// For convT: Y = input, X = grad_out (the larger map), giving the
+// (Ci,R,S,Co)-physical transposed-conv weight grad.
+//
+// GEMM view per (r,s): (B x A) = Y^T @ X_patch, reduced over all N*HO*WO
+// pixels — a huge-K GEMM, split over pixel slabs (grid.z) with fp32
+// atomicAdd into a workspace (deterministic mode: splitP=1).
+//
+// Both operands are pixel-major in memory (NHWC), so global loads are
+// contiguous 16B channel runs; tiles are TRANSPOSED while staging into LDS
+// ([channel][pixel], odd row pitch so the b128 fragment reads are spread
+// across banks), which makes every MFMA fragment read a contiguous
+// ds_read_b128 along the pixel (reduction) axis.
+
+#include "common.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+namespace {
+
+constexpr int BT = 64;     // channel tile (both B and A side)
+constexpr int PCH = 64;    // pixels per staging chunk
+constexpr int THREADS = 256;
+constexpr int PIT = PCH + 8;          // LDS row pitch in elements (16B-aligned)
+constexpr int ROWB = PIT * 2;         // bytes per LDS row
+
+__global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
+    const __bf16* __restrict__ Y,  // (N, HO, WO, B)
+    const __bf16* __restrict__ X,  // (N, H, W, A)
+    float* __restrict__ ws,        // (B, R, S, A) fp32, pre-zeroed
+    int Nb, int HO, int WO, int B, int H, int W, int A, int R, int S,
+    int STRIDE, int PAD, int p_per_slab) {
+  __shared__ __align__(16) char lds[2 * BT * ROWB + PCH * 12];
+  char* yt = lds;                       // [BT rows (b)][PIT pix]
+  char* xt = lds + BT * ROWB;           // [BT rows (a)][PIT pix]
+  int* pix_off = reinterpret_cast<int*>(lds + 2 * BT * ROWB);  // X base offset
+  short* pix_hi = reinterpret_cast<short*>(pix_off + PCH);
+  short* pix_wi = pix_hi + PCH;
+
+  const int at_blocks = (A + BT - 1) / BT;
+  const int b0 = (blockIdx.x / at_blocks) * BT;
+  const int a0 = (blockIdx.x % at_blocks) * BT;
+  const int r = blockIdx.y / S;
+  const int s = blockIdx.y % S;
+  const int p_begin = blockIdx.z * p_per_slab;
+  const int p_total = Nb * HO * WO;
+  const int p_end = min(p_begin + p_per_slab, p_total);
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = (wid >> 1) * 32;   // wave row (b) base
+  const int wn = (wid & 1) * 32;    // wave col (a) base
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int p0 = p_begin; p0 < p_end; p0 += PCH) {
+    // pixel meta for this chunk
+    if (tid < PCH) {
+      const int pix = p0 + tid;
+      if (pix < p_end) {
+        const int n = pix / (HO * WO);
+        const int rem = pix - n * (HO * WO);
+        const int ho = rem / WO;
+        const int wo = rem - ho * WO;
+        pix_off[tid] = n * H * W * A;
+        pix_hi[tid] = (short)(ho * STRIDE - PAD + r);
+        pix_wi[tid] = (short)(wo * STRIDE - PAD + s);
+      } else {
+        pix_off[tid] = 0;
+        pix_hi[tid] = (short)-30000;
+        pix_wi[tid] = (short)-30000;
+      }
+    }
+    __syncthreads();
+
+    // stage Y^T: 64 pix x 8 chunks of 8 channels
+#pragma unroll
+    for (int it = 0; it < (PCH * 8) / THREADS; ++it) {
+      const int slot = it * THREADS + tid;
+      const int pix_l = slot >> 3;
+      const int ch0 = (slot & 7) * 8;
+      const int pix = p0 + pix_l;
+      bf16x8 v = {};
+      if (pix < p_end && b0 + ch0 < B) {
+        const __bf16* src = Y + (long)pix * B + b0 + ch0;
+        if (b0 + ch0 + 8 <= B) {
+          v = *reinterpret_cast<const bf16x8*>(src);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (b0 + ch0 + j < B) v[j] = src[j];
+        }
+      }
+      __bf16* dst = reinterpret_cast<__bf16*>(yt);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[(ch0 + j) * PIT + pix_l] = v[j];
+    }
+    // stage X^T (patch)
+#pragma unroll
+    for (int it = 0; it < (PCH * 8) / THREADS; ++it) {
+      const int slot = it * THREADS + tid;
+      const int pix_l = slot >> 3;
+      const int ch0 = (slot & 7) * 8;
+      const int hi = pix_hi[pix_l];
+      const int wi = pix_wi[pix_l];
+      bf16x8 v = {};
+      if (hi >= 0 && hi < H && wi >= 0 && wi < W && a0 + ch0 < A) {
+        const __bf16* src =
+            X + (long)pix_off[pix_l] + ((long)hi * W + wi) * A + a0 + ch0;
+        if (a0 + ch0 + 8 <= A) {
+          v = *reinterpret_cast<const bf16x8*>(src);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (a0 + ch0 + j < A) v[j] = src[j];
+        }
+      }
+      __bf16* dst = reinterpret_cast<__bf16*>(xt);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dst[(ch0 + j) * PIT + pix_l] = v[j];
+    }
+    __syncthreads();
+
+    // MFMA: 2 K-steps of 32 pixels
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int pb = (kk * 32 + (lane >> 4) * 8) * 2;  // byte offset along pix
+      bf16x8 a_frag[2], b_frag[2];
+#pragma unroll
+      for (int f = 0; f < 2; ++f) {
+        const int brow = wm + f * 16 + (lane & 15);
+        a_frag[f] = *reinterpret_cast<const bf16x8*>(yt + brow * ROWB + pb);
+        const int arow = wn + f * 16 + (lane & 15);
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(xt + arow * ROWB + pb);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: atomic accumulate into the fp32 workspace
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int a = a0 + wn + j * 16 + (lane & 15);
+    if (a >= A) continue;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+#pragma unroll
+      for (int v = 0; v < 4; ++v) {
+        const int b = b0 + wm + i * 16 + (lane >> 4) * 4 + v;
+        if (b < B) {
+          atomicAdd(&ws[(((long)b * R + r) * S + s) * A + a], acc[i][j][v]);
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// Y: channels_last (N,B,HO,WO) bf16; X: channels_last (N,A,H,W) bf16.
+// Returns dW workspace (B, R, S, A) fp32. splitp<=0 -> auto.
+torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
+                                long S, long stride, long pad, long splitp) {
+  TORCH_CHECK(Y.is_cuda() && Y.scalar_type() == torch::kBFloat16 &&
+                  Y.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "Y must be bf16 channels_last GPU");
+  TORCH_CHECK(X.is_cuda() && X.scalar_type() == torch::kBFloat16 &&
+                  X.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "X must be bf16 channels_last GPU");
+  const int Nb = Y.size(0), B = Y.size(1), HO = Y.size(2), WO = Y.size(3);
+  const int A = X.size(1), H = X.size(2), W = X.size(3);
+  TORCH_CHECK(X.size(0) == Nb, "batch mismatch");
+
+  auto ws = torch::zeros({B, (long)R, (long)S, A},
+                         Y.options().dtype(torch::kFloat32));
+
+  const int bt = ceil_div(B, BT), at = ceil_div(A, BT);
+  const int p_total = Nb * HO * WO;
+  int sp = (int)splitp;
+  if (sp <= 0) {
+    sp = std::max(1, 1024 / std::max(1, bt * at * (int)R * (int)S));
+    sp = std::min(sp, ceil_div(p_total, PCH));
+  }
+  const int p_per_slab =
+      ceil_div(ceil_div(p_total, sp), PCH) * PCH;  // chunk-aligned
+  sp = ceil_div(p_total, p_per_slab);
+
+  dim3 grid(bt * at, (int)(R * S), sp);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(THREADS), 0, stream,
+                     reinterpret_cast<const __bf16*>(Y.data_ptr()),
+                     reinterpret_cast<const __bf16*>(X.data_ptr()),
+                     ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
+                     (int)S, (int)stride, (int)pad, p_per_slab);
+  return ws;
+}
