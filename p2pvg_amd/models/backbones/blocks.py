@@ -13,6 +13,7 @@ import torch
 import torch.nn as nn
 
 from ...ops.conv import Conv2d, ConvTranspose2d
+from ...ops.norm import BatchNorm2d
 
 
 class dcgan_conv(nn.Module):
@@ -22,7 +23,7 @@ class dcgan_conv(nn.Module):
         super().__init__()
         self.main = nn.Sequential(
             Conv2d(nin, nout, 4, 2, 1),
-            nn.BatchNorm2d(nout),
+            BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
         )
 
@@ -37,7 +38,7 @@ class dcgan_upconv(nn.Module):
         super().__init__()
         self.main = nn.Sequential(
             ConvTranspose2d(nin, nout, 4, 2, 1),
-            nn.BatchNorm2d(nout),
+            BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
         )
 
@@ -52,7 +53,7 @@ class vgg_layer(nn.Module):
         super().__init__()
         self.main = nn.Sequential(
             Conv2d(nin, nout, 3, 1, 1),
-            nn.BatchNorm2d(nout),
+            BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
         )
 

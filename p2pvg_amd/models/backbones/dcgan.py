@@ -11,6 +11,7 @@ import torch
 import torch.nn as nn
 
 from ...ops.conv import Conv2d, ConvTranspose2d
+from ...ops.norm import BatchNorm2d
 
 from .blocks import dcgan_conv, dcgan_upconv
 
@@ -26,7 +27,7 @@ class Encoder64(nn.Module):
         self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x4x4
         self.c5 = nn.Sequential(                  # -> dim x1x1
             Conv2d(nf * 8, dim, 4, 1, 0),
-            nn.BatchNorm2d(dim),
+            BatchNorm2d(dim),
             nn.Tanh(),
         )
 
@@ -46,7 +47,7 @@ class Decoder64(nn.Module):
         nf = 64
         self.upc1 = nn.Sequential(
             ConvTranspose2d(dim, nf * 8, 4, 1, 0),
-            nn.BatchNorm2d(nf * 8),
+            BatchNorm2d(nf * 8),
             nn.LeakyReLU(0.2, inplace=True),
         )
         self.upc2 = dcgan_upconv(nf * 8 * 2, nf * 4)
@@ -80,7 +81,7 @@ class Encoder128(nn.Module):
         self.c5 = dcgan_conv(nf * 8, nf * 8)      # -> 512x4
         self.c6 = nn.Sequential(
             Conv2d(nf * 8, dim, 4, 1, 0),
-            nn.BatchNorm2d(dim),
+            BatchNorm2d(dim),
             nn.Tanh(),
         )
 
@@ -101,7 +102,7 @@ class Decoder128(nn.Module):
         nf = 64
         self.upc1 = nn.Sequential(
             ConvTranspose2d(dim, nf * 8, 4, 1, 0),
-            nn.BatchNorm2d(nf * 8),
+            BatchNorm2d(nf * 8),
             nn.LeakyReLU(0.2, inplace=True),
         )
         self.upc2 = dcgan_upconv(nf * 8 * 2, nf * 8)

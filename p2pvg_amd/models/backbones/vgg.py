@@ -11,6 +11,7 @@ import torch
 import torch.nn as nn
 
 from ...ops.conv import Conv2d, ConvTranspose2d
+from ...ops.norm import BatchNorm2d
 
 from .blocks import vgg_layer
 
@@ -28,7 +29,7 @@ class Encoder64(nn.Module):
             vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
         self.c5 = nn.Sequential(
-            Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
+            Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
         )
         self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
 
@@ -47,7 +48,7 @@ class Decoder64(nn.Module):
         self.dim = dim
         self.upc1 = nn.Sequential(
             ConvTranspose2d(dim, 512, 4, 1, 0),
-            nn.BatchNorm2d(512),
+            BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
         self.upc2 = nn.Sequential(
@@ -89,7 +90,7 @@ class Encoder128(nn.Module):
             vgg_layer(512, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
         self.c6 = nn.Sequential(
-            Conv2d(512, dim, 4, 1, 0), nn.BatchNorm2d(dim), nn.Tanh()
+            Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
         )
         self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
 
@@ -109,7 +110,7 @@ class Decoder128(nn.Module):
         self.dim = dim
         self.upc1 = nn.Sequential(
             ConvTranspose2d(dim, 512, 4, 1, 0),
-            nn.BatchNorm2d(512),
+            BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
         self.upc2 = nn.Sequential(
