@@ -2,25 +2,28 @@
 //
 // Covers the conv workloads of SURVEY §2.6 K1-K3 (dispatch sites
 // reference models/vgg_64.py:8 k3s1p1, models/dcgan_64.py:8 k4s2p1,
-// encoder tails k4s1p0, decoder ConvTranspose k4s2p1):
+// decoder ConvTranspose k4s2p1):
 //   out[n,yo,xo,k] = act(bias[k] + sum_{r,s,c} in[n, ho*S-P+r, wo*S-P+s, c]
 //                                              * w[k,r,s,c])
-// where (yo,xo) = (ho*OYS+OY0, wo*OXS+OX0): OYS=1 is plain convolution;
-// OYS=2 with a per-parity weight slice realizes fractionally-strided
-// convolution (ConvTranspose fwd / stride-2 dgrad) as 4 dense sub-problems —
-// the xGMI-free, transpose-free CDNA4 formulation of K2/K3.
 //
-// GEMM view: M = N*HO*WO output pixels, Ndim = K output channels,
-// Kdim = R*S*C, iterated as (r,s) outer x 64-wide c-chunks inner so the
-// im2col gather of one chunk is a CONTIGUOUS 128-byte run of the NHWC input
-// per pixel (c fastest) — staged straight into LDS.
+// Three staging modes share one MFMA core:
+// - dense (r,s)-outer / 64-wide c-chunk-inner: the im2col gather of a chunk
+//   is a CONTIGUOUS 128-byte NHWC run per pixel.
+// - RSCLIN (C < 32, e.g. the nc=1/3 first layers): chunks walk the rsc index
+//   linearly so a chunk packs many taps — no zero-padding waste; weights stay
+//   contiguous, input gathers go per-element.
+// - FRAC (fractionally-strided): grid.z = stride^2 parities, each with its
+//   own tap map (r,s taps valid for that output parity) and scatter offsets;
+//   realizes ConvTranspose fwd / stride-2 dgrad as dense stride-1
+//   sub-convolutions in ONE launch — no dilated intermediates, no transposes,
+//   no host-side weight slicing.
 //
-// Tiling: 128x128 block tile (BM pixels x BN channels), 4 waves as 2x2 of
-// 64x64 wave tiles, v_mfma_f32_16x16x32_bf16 with fp32 accumulation, LDS
-// tiles XOR-swizzled (byte ^= (row&7)<<4) so the ds_read_b128 fragment reads
-// are <=2-way bank conflicted (guide §6 G4 / T2). Epilogue fuses bias +
-// activation (none/LeakyReLU(0.2)/Tanh/Sigmoid) — the K5 fusion.
-// blockIdx.x is XCD-swizzled over the M dimension (T1).
+// Tiling: template (BM x BN) block tile over (pixels x channels), 4 waves as
+// 2x2, v_mfma_f32_16x16x32_bf16 with fp32 accumulation, LDS tiles
+// XOR-swizzled (byte ^= (row&7)<<4) for <=2-way ds_read_b128 conflicts
+// (guide §6 G4 / T2). Epilogue fuses bias + activation (K5) and optional
+// per-channel sum/sumsq accumulation for downstream BatchNorm (K4).
+// blockIdx.x is XCD-swizzled over M (T1).
 
 #include "common.h"
 
@@ -29,14 +32,11 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
 
-constexpr int BM = 128;
-constexpr int BN = 128;
-constexpr int BK = 64;              // c-chunk (bf16 elements)
+constexpr int BK = 64;              // chunk width (bf16 elements)
 constexpr int THREADS = 256;
 constexpr int ROW_BYTES = BK * 2;   // 128 B per LDS tile row
 
 __device__ __forceinline__ int swz(int row, int cb) {
-  // byte offset into a [rows][BK] bf16 tile with the (row&7)<<4 XOR swizzle
   return row * ROW_BYTES + (cb ^ ((row & 7) << 4));
 }
 
@@ -49,20 +49,28 @@ __device__ __forceinline__ float activate(float v, int act) {
   }
 }
 
-struct ScatterSpec {
-  int OH, OW;    // full output spatial dims
-  int oys, oy0, ox0;  // y = ho*oys + oy0, x = wo*oys + ox0
+struct ConvArgs {
+  int Nb, H, W, C, K;
+  int HO, WO;           // compact output grid (per parity when FRAC)
+  int OH, OW;           // full output spatial dims
+  int act;
+  int mblocks;
+  int wk;               // full weight kernel size (for FRAC weight indexing)
+  // per-parity tap maps and scatter (index [parity][tap]); parity 0 used
+  // for the non-FRAC case with identity maps.
+  int rmap[4][4];
+  int smap[4][4];
+  int padh[4], padw[4];
+  int oy0[4], ox0[4], oys;
+  float* stats;         // optional (2,K) sum/sumsq accumulation, or null
 };
 
-// in:  (N, H, W, C) bf16   w: (K, R, S, C) bf16   bias: (K) f32 or null
-// out: (N, OH, OW, K) bf16, written at the scattered (y,x) positions
-template <int KSIZE, int STRIDE>
+// KSIZE: compact kernel size (taps per axis). FRAC: parity mode. RSCLIN:
+// rsc-linear chunking for small C.
+template <int BM, int BN, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
 __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ w,
-    const float* __restrict__ bias, __bf16* __restrict__ out,
-    int Nb, int H, int W, int C, int K, int HO, int WO, int PADH, int PADW,
-    int act, int mblocks, ScatterSpec sc) {
-  // LDS: A tile (BM x BK) + B tile (BN x BK), single-buffered, + pixel meta
+    const float* __restrict__ bias, __bf16* __restrict__ out, ConvArgs a) {
   __shared__ __align__(16) char lds[(BM + BN) * ROW_BYTES + BM * 16];
   char* a_lds = lds;
   char* b_lds = lds + BM * ROW_BYTES;
@@ -71,184 +79,243 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
   short* pix_hi = reinterpret_cast<short*>(pix_off + BM);
   short* pix_wi = pix_hi + BM;
 
-  // XCD-aware remap of the M dimension (bijective variant, guide §5)
+  const int par = FRAC ? blockIdx.z : 0;
+
+  // XCD-aware bijective remap of the M dimension
   int bm_lin = blockIdx.x;
   {
     const int nxcd = 8;
-    const int q = mblocks / nxcd, r = mblocks % nxcd;
+    const int q = a.mblocks / nxcd, r = a.mblocks % nxcd;
     const int xcd = bm_lin % nxcd, idx = bm_lin / nxcd;
     bm_lin = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
   const int m0 = bm_lin * BM;
   const int k0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
-  const int M = Nb * HO * WO;
+  const int M = a.Nb * a.HO * a.WO;
 
-  // pixel meta
   for (int i = tid; i < BM; i += THREADS) {
     const int pix = m0 + i;
     if (pix < M) {
-      const int n = pix / (HO * WO);
-      const int rem = pix - n * (HO * WO);
-      const int ho = rem / WO;
-      const int wo = rem - ho * WO;
-      pix_off[i] = n * H * W * C;
-      pix_hi[i] = (short)(ho * STRIDE - PADH);
-      pix_wi[i] = (short)(wo * STRIDE - PADW);
+      const int n = pix / (a.HO * a.WO);
+      const int rem = pix - n * (a.HO * a.WO);
+      const int ho = rem / a.WO;
+      const int wo = rem - ho * a.WO;
+      pix_off[i] = n * a.H * a.W * a.C;
+      pix_hi[i] = (short)(ho * STRIDE - a.padh[par]);
+      pix_wi[i] = (short)(wo * STRIDE - a.padw[par]);
       pix_out[i] =
-          (((long)n * sc.OH + (ho * sc.oys + sc.oy0)) * sc.OW +
-           (wo * sc.oys + sc.ox0)) *
-          K;
+          (((long)n * a.OH + (ho * a.oys + a.oy0[par])) * a.OW +
+           (wo * a.oys + a.ox0[par])) *
+          a.K;
     } else {
       pix_off[i] = 0;
-      pix_hi[i] = (short)-30000;  // always out of bounds -> zero rows
+      pix_hi[i] = (short)-30000;
       pix_wi[i] = (short)-30000;
       pix_out[i] = -1;
     }
   }
   __syncthreads();
 
-  const int wid = tid >> 6;         // wave 0..3
+  const int wid = tid >> 6;
   const int lane = tid & 63;
-  const int wm = (wid >> 1) * 64;   // wave row base in tile
-  const int wn = (wid & 1) * 64;    // wave col base in tile
+  constexpr int FM = BM / 32;        // M fragments per wave (wave tile BM/2)
+  constexpr int FN = BN / 32;        // N fragments per wave
+  const int wm = (wid >> 1) * (BM / 2);
+  const int wn = (wid & 1) * (BN / 2);
 
-  f32x4 acc[4][4];
+  f32x4 acc[FM][FN];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < FM; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const bool cvec = (C % 8) == 0;
+  const bool cvec = (a.C % 8) == 0;
+  const int RSC = KSIZE * KSIZE * a.C;
+  // chunk loop bounds
+  const int n_outer = RSCLIN ? 1 : KSIZE * KSIZE;
+  const int n_inner = RSCLIN ? (RSC + BK - 1) / BK : (a.C + BK - 1) / BK;
 
-  for (int r = 0; r < KSIZE; ++r) {
-    for (int s = 0; s < KSIZE; ++s) {
-      for (int c0 = 0; c0 < C; c0 += BK) {
-        // ---- stage A (input patches) ----
-        // 128 rows x 8 slots of 16B; 1024 slots over 256 threads
+  for (int oidx = 0; oidx < n_outer; ++oidx) {
+    const int ro = oidx / KSIZE, so = oidx % KSIZE;
+    for (int ci = 0; ci < n_inner; ++ci) {
+      const int c0 = ci * BK;
+      // ---- stage A ----
 #pragma unroll
-        for (int it = 0; it < (BM * 8) / THREADS; ++it) {
-          const int slot = it * THREADS + tid;
-          const int row = slot >> 3;
-          const int cb = (slot & 7) * 16;         // byte col
-          const int c = c0 + (cb >> 1);           // element col base
-          const int hi = pix_hi[row] + r;
-          const int wi = pix_wi[row] + s;
-          bf16x8 v = {};
-          if (hi >= 0 && hi < H && wi >= 0 && wi < W && c < C) {
-            const __bf16* src = in + (long)pix_off[row] + ((long)hi * W + wi) * C + c;
-            if (cvec && c + 8 <= C) {
-              v = *reinterpret_cast<const bf16x8*>(src);
-            } else {
+      for (int it = 0; it < (BM * 8) / THREADS; ++it) {
+        const int slot = it * THREADS + tid;
+        const int row = slot >> 3;
+        const int cb = (slot & 7) * 16;
+        bf16x8 v = {};
+        if (RSCLIN) {
 #pragma unroll
-              for (int j = 0; j < 8; ++j)
-                if (c + j < C) v[j] = src[j];
+          for (int j = 0; j < 8; ++j) {
+            const int q = c0 + (cb >> 1) + j;
+            if (q < RSC) {
+              const int rs = q / a.C;
+              const int c = q - rs * a.C;
+              const int r = rs / KSIZE, s = rs % KSIZE;
+              const int hi = pix_hi[row] + r;
+              const int wi = pix_wi[row] + s;
+              if (hi >= 0 && hi < a.H && wi >= 0 && wi < a.W)
+                v[j] = in[(long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c];
             }
           }
-          *reinterpret_cast<bf16x8*>(a_lds + swz(row, cb)) = v;
-        }
-        // ---- stage B (weights) ----
-#pragma unroll
-        for (int it = 0; it < (BN * 8) / THREADS; ++it) {
-          const int slot = it * THREADS + tid;
-          const int row = slot >> 3;              // out-channel within tile
-          const int cb = (slot & 7) * 16;
+        } else {
           const int c = c0 + (cb >> 1);
-          const int k = k0 + row;
-          bf16x8 v = {};
-          if (k < K && c < C) {
-            const __bf16* src = w + (((long)k * KSIZE + r) * KSIZE + s) * C + c;
-            if (cvec && c + 8 <= C) {
+          const int hi = pix_hi[row] + ro;
+          const int wi = pix_wi[row] + so;
+          if (hi >= 0 && hi < a.H && wi >= 0 && wi < a.W && c < a.C) {
+            const __bf16* src =
+                in + (long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c;
+            if (cvec && c + 8 <= a.C) {
               v = *reinterpret_cast<const bf16x8*>(src);
             } else {
 #pragma unroll
               for (int j = 0; j < 8; ++j)
-                if (c + j < C) v[j] = src[j];
+                if (c + j < a.C) v[j] = src[j];
             }
           }
-          *reinterpret_cast<bf16x8*>(b_lds + swz(row, cb)) = v;
         }
-        __syncthreads();
-
-        // ---- MFMA over the 64-wide chunk (2 x K=32 steps) ----
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
-          const int cb = kk * 64 + ((lane >> 4) * 16);
-          bf16x8 a_frag[4], b_frag[4];
-#pragma unroll
-          for (int f = 0; f < 4; ++f) {
-            const int arow = wm + f * 16 + (lane & 15);
-            a_frag[f] = *reinterpret_cast<const bf16x8*>(a_lds + swz(arow, cb));
-            const int brow = wn + f * 16 + (lane & 15);
-            b_frag[f] = *reinterpret_cast<const bf16x8*>(b_lds + swz(brow, cb));
-          }
-#pragma unroll
-          for (int i = 0; i < 4; ++i)
-#pragma unroll
-            for (int j = 0; j < 4; ++j)
-              acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
-        }
-        __syncthreads();
+        *reinterpret_cast<bf16x8*>(a_lds + swz(row, cb)) = v;
       }
+      // ---- stage B (weights) ----
+#pragma unroll
+      for (int it = 0; it < (BN * 8) / THREADS; ++it) {
+        const int slot = it * THREADS + tid;
+        const int row = slot >> 3;
+        const int cb = (slot & 7) * 16;
+        const int k = k0 + row;
+        bf16x8 v = {};
+        if (k < a.K) {
+          if (RSCLIN) {
+            // weight rows are rsc-contiguous; remap taps only when FRAC
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              const int q = c0 + (cb >> 1) + j;
+              if (q < RSC) {
+                int off;
+                if (FRAC) {
+                  const int rs = q / a.C;
+                  const int c = q - rs * a.C;
+                  const int r = a.rmap[par][rs / KSIZE];
+                  const int s = a.smap[par][rs % KSIZE];
+                  off = ((k * a.wk + r) * a.wk + s) * a.C + c;
+                } else {
+                  off = k * RSC + q;
+                }
+                v[j] = w[(long)off];
+              }
+            }
+          } else {
+            const int c = c0 + (cb >> 1);
+            const int r = FRAC ? a.rmap[par][ro] : ro;
+            const int s = FRAC ? a.smap[par][so] : so;
+            if (c < a.C) {
+              const __bf16* src =
+                  w + (((long)k * a.wk + r) * a.wk + s) * a.C + c;
+              if (cvec && c + 8 <= a.C) {
+                v = *reinterpret_cast<const bf16x8*>(src);
+              } else {
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                  if (c + j < a.C) v[j] = src[j];
+              }
+            }
+          }
+        }
+        *reinterpret_cast<bf16x8*>(b_lds + swz(row, cb)) = v;
+      }
+      __syncthreads();
+
+      // ---- MFMA over the 64-wide chunk (2 x K=32 steps) ----
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int cb = kk * 64 + ((lane >> 4) * 16);
+        bf16x8 a_frag[FM], b_frag[FN];
+#pragma unroll
+        for (int f = 0; f < FM; ++f) {
+          const int arow = wm + f * 16 + (lane & 15);
+          a_frag[f] = *reinterpret_cast<const bf16x8*>(a_lds + swz(arow, cb));
+        }
+#pragma unroll
+        for (int f = 0; f < FN; ++f) {
+          const int brow = wn + f * 16 + (lane & 15);
+          b_frag[f] = *reinterpret_cast<const bf16x8*>(b_lds + swz(brow, cb));
+        }
+#pragma unroll
+        for (int i = 0; i < FM; ++i)
+#pragma unroll
+          for (int j = 0; j < FN; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+      }
+      __syncthreads();
     }
   }
 
-  // ---- epilogue: bias + activation, bf16 store ----
+  // ---- epilogue ----
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < FN; ++j) {
     const int col = k0 + wn + j * 16 + (lane & 15);
-    if (col >= K) continue;
+    if (col >= a.K) continue;
     const float bv = bias != nullptr ? bias[col] : 0.f;
+    float csum = 0.f, csq = 0.f;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < FM; ++i) {
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
         const int lrow = wm + i * 16 + (lane >> 4) * 4 + v;
         const long ooff = pix_out[lrow];
         if (ooff >= 0) {
-          const float val = activate(acc[i][j][v] + bv, act);
+          const float val = activate(acc[i][j][v] + bv, a.act);
           out[ooff + col] = (__bf16)val;
+          csum += val;
+          csq += val * val;
         }
       }
+    }
+    if (a.stats != nullptr) {
+      atomicAdd(&a.stats[col], csum);
+      atomicAdd(&a.stats[a.K + col], csq);
     }
   }
 }
 
-template <int KSIZE, int STRIDE>
-void launch_fwd(const torch::Tensor& in, const torch::Tensor& w,
+template <int BM, int BN, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
+void launch_one(const torch::Tensor& in, const torch::Tensor& w,
                 const c10::optional<torch::Tensor>& bias, torch::Tensor& out,
-                int Nb, int H, int W, int C, int K, int HO, int WO, int PADH,
-                int PADW, int act, ScatterSpec sc) {
-  const int M = Nb * HO * WO;
-  const int mblocks = ceil_div(M, BM);
-  dim3 grid(mblocks, ceil_div(K, BN));
+                ConvArgs& a, int nz) {
+  const int M = a.Nb * a.HO * a.WO;
+  a.mblocks = ceil_div(M, BM);
+  dim3 grid(a.mblocks, ceil_div(a.K, BN), nz);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL((conv2d_nhwc_fwd_kernel<KSIZE, STRIDE>), grid,
-                     dim3(THREADS), 0, stream,
-                     reinterpret_cast<const __bf16*>(in.data_ptr()),
-                     reinterpret_cast<const __bf16*>(w.data_ptr()),
-                     bias.has_value() ? bias->data_ptr<float>() : nullptr,
-                     reinterpret_cast<__bf16*>(out.data_ptr()), Nb, H, W, C, K,
-                     HO, WO, PADH, PADW, act, mblocks, sc);
+  hipLaunchKernelGGL(
+      (conv2d_nhwc_fwd_kernel<BM, BN, KSIZE, STRIDE, FRAC, RSCLIN>), grid,
+      dim3(THREADS), 0, stream,
+      reinterpret_cast<const __bf16*>(in.data_ptr()),
+      reinterpret_cast<const __bf16*>(w.data_ptr()),
+      bias.has_value() ? bias->data_ptr<float>() : nullptr,
+      reinterpret_cast<__bf16*>(out.data_ptr()), a);
 }
 
-void dispatch_fwd(const torch::Tensor& in, const torch::Tensor& w,
-                  const c10::optional<torch::Tensor>& bias, torch::Tensor& out,
-                  int Nb, int H, int W, int C, int K, int HO, int WO, int R,
-                  int stride, int PADH, int PADW, int act, ScatterSpec sc) {
-  if (R == 3 && stride == 1) {
-    launch_fwd<3, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
-  } else if (R == 4 && stride == 2) {
-    launch_fwd<4, 2>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
-  } else if (R == 4 && stride == 1) {
-    launch_fwd<4, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
-  } else if (R == 2 && stride == 1) {
-    launch_fwd<2, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
-  } else if (R == 1 && stride == 1) {
-    launch_fwd<1, 1>(in, w, bias, out, Nb, H, W, C, K, HO, WO, PADH, PADW, act, sc);
+// choose tile by problem size, then kernel geometry
+template <int KSIZE, int STRIDE, bool FRAC>
+void dispatch_tile(const torch::Tensor& in, const torch::Tensor& w,
+                   const c10::optional<torch::Tensor>& bias,
+                   torch::Tensor& out, ConvArgs& a, int nz) {
+  const bool rsclin = !FRAC && a.C < 32 && KSIZE > 1;
+  const int M = a.Nb * a.HO * a.WO;
+  const bool small = (long)ceil_div(M, 128) * ceil_div(a.K, 128) < 160;
+  if (rsclin) {
+    if (small && a.K <= 64)
+      launch_one<64, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+    else
+      launch_one<128, 128, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+  } else if (small) {
+    launch_one<64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   } else {
-    TORCH_CHECK(false, "unsupported conv geometry: k=", R, " stride=", stride);
+    launch_one<128, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   }
 }
 
@@ -263,9 +330,12 @@ void check_nhwc_bf16(const torch::Tensor& t, const char* name) {
 
 // in: channels_last (N,C,H,W) bf16; w: channels_last (K,C,R,S) bf16;
 // bias: (K) fp32 optional. act: 0 none, 1 leaky(0.2), 2 tanh, 3 sigmoid.
+// stats: optional (2,K) fp32 ZEROED buffer accumulating sum/sumsq of the
+// activated output per channel (for fused BatchNorm statistics).
 torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
                               c10::optional<torch::Tensor> bias, long stride,
-                              long pad, long act) {
+                              long pad, long act,
+                              c10::optional<torch::Tensor> stats) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
@@ -281,34 +351,106 @@ torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
     CHECK_INPUT(bias.value());
     TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be fp32");
   }
-  ScatterSpec sc{HO, WO, 1, 0, 0};
-  dispatch_fwd(in, w, bias, out, Nb, H, W, C, K, HO, WO, R, (int)stride,
-               (int)pad, (int)pad, (int)act, sc);
+  ConvArgs a{};
+  a.Nb = Nb; a.H = H; a.W = W; a.C = C; a.K = K;
+  a.HO = HO; a.WO = WO; a.OH = HO; a.OW = WO;
+  a.act = (int)act; a.wk = R; a.oys = 1;
+  a.padh[0] = (int)pad; a.padw[0] = (int)pad;
+  a.stats = stats.has_value() ? stats->data_ptr<float>() : nullptr;
+
+  if (R == 3 && stride == 1) {
+    dispatch_tile<3, 1, false>(in, w, bias, out, a, 1);
+  } else if (R == 4 && stride == 2) {
+    dispatch_tile<4, 2, false>(in, w, bias, out, a, 1);
+  } else if (R == 4 && stride == 1) {
+    dispatch_tile<4, 1, false>(in, w, bias, out, a, 1);
+  } else if (R == 2 && stride == 1) {
+    dispatch_tile<2, 1, false>(in, w, bias, out, a, 1);
+  } else if (R == 1 && stride == 1) {
+    dispatch_tile<1, 1, false>(in, w, bias, out, a, 1);
+  } else {
+    TORCH_CHECK(false, "unsupported conv geometry: k=", R, " stride=", stride);
+  }
   return out;
 }
 
-// Scatter variant: computes a stride-1 conv of `in` with `w` on a compact
-// HOxWO grid and writes results at out[:, :, oy0::oys, ox0::oxs]. `out` is
-// preallocated by the caller (one call per parity). ipad is the implicit
-// input padding of the compact problem.
-void conv2d_nhwc_fwd_scatter(torch::Tensor in, torch::Tensor w,
-                             c10::optional<torch::Tensor> bias,
-                             torch::Tensor out, long ipad_h, long ipad_w,
-                             long oys, long oy0, long ox0, long act) {
+// Fractionally-strided convolution (ConvTranspose fwd / stride-s dgrad):
+// out[n, y, x, k] = sum over taps with y = ho*s + py etc. One launch,
+// grid.z = s^2 parities. `w` is the FULL (K, C, wk, wk) channels_last weight
+// (already arranged so dim0 = output channels); taps are remapped in-kernel.
+// up_pad is the fractional-conv padding (the original conv's pad).
+torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
+                                     c10::optional<torch::Tensor> bias,
+                                     long up_stride, long up_pad, long OH,
+                                     long OW, long act) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
-  check_nhwc_bf16(out, "out");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
-  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  const int K = w.size(0), WK = w.size(2);
   TORCH_CHECK(w.size(1) == C, "channel mismatch");
-  TORCH_CHECK(R == S, "square kernels only");
-  const int OH = out.size(2), OW = out.size(3);
-  // compact grid dims from the scatter spec
-  const int HO = (OH - 1 - (int)oy0) / (int)oys + 1;
-  const int WO = (OW - 1 - (int)ox0) / (int)oys + 1;
-  TORCH_CHECK(out.size(0) == Nb && out.size(1) == K, "bad out shape");
+  TORCH_CHECK(up_stride == 2, "fracstride: stride 2 only");
+  const int st = (int)up_stride, pad = (int)up_pad;
 
-  ScatterSpec sc{OH, OW, (int)oys, (int)oy0, (int)ox0};
-  dispatch_fwd(in, w, bias, out, Nb, H, W, C, K, HO, WO, R, 1, (int)ipad_h,
-               (int)ipad_w, (int)act, sc);
+  auto out = torch::empty({Nb, K, OH, OW},
+                          in.options().memory_format(at::MemoryFormat::ChannelsLast));
+  if (bias.has_value()) {
+    CHECK_INPUT(bias.value());
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be fp32");
+  }
+
+  ConvArgs a{};
+  a.Nb = Nb; a.H = H; a.W = W; a.C = C; a.K = K;
+  a.OH = (int)OH; a.OW = (int)OW;
+  a.act = (int)act; a.wk = WK; a.oys = st;
+  a.stats = nullptr;
+
+  // taps per parity: r with (p + pad - r) % st == 0, offset (p+pad-r)/st
+  int ktaps = -1;
+  int HOc = -1, WOc = -1;
+  for (int p = 0; p < st; ++p) {
+    int taps[8][2], nt = 0;
+    for (int r = 0; r < WK; ++r) {
+      if ((p + pad - r) % st == 0) {
+        taps[nt][0] = r;
+        taps[nt][1] = (p + pad - r) / st;
+        ++nt;
+      }
+    }
+    // sort by offset ascending
+    for (int i = 0; i < nt; ++i)
+      for (int j = i + 1; j < nt; ++j)
+        if (taps[j][1] < taps[i][1]) {
+          std::swap(taps[i][0], taps[j][0]);
+          std::swap(taps[i][1], taps[j][1]);
+        }
+    const int ipad = -taps[0][1];
+    for (int i = 0; i < nt; ++i)
+      TORCH_CHECK(taps[i][1] == i - ipad, "non-consecutive tap offsets");
+    if (ktaps < 0) ktaps = nt;
+    TORCH_CHECK(nt == ktaps, "tap count varies across parities (pad=", pad,
+                " k=", WK, ")");
+    // the same axis data serves y-parity rows and x-parity columns
+    for (int px = 0; px < st; ++px) {
+      const int pi = p * st + px;
+      for (int i = 0; i < nt; ++i) a.rmap[pi][i] = taps[i][0];
+      a.padh[pi] = ipad;
+      a.oy0[pi] = p;
+    }
+    for (int py = 0; py < st; ++py) {
+      const int pi = py * st + p;
+      for (int i = 0; i < nt; ++i) a.smap[pi][i] = taps[i][0];
+      a.padw[pi] = ipad;
+      a.ox0[pi] = p;
+    }
+    // compact grid dims are the same for all parities when OH even
+    const int hoc = ((int)OH - 1 - p) / st + 1;
+    if (HOc < 0) HOc = hoc; else TORCH_CHECK(hoc == HOc, "uneven parity grid");
+    const int woc = ((int)OW - 1 - p) / st + 1;
+    if (WOc < 0) WOc = woc; else TORCH_CHECK(woc == WOc, "uneven parity grid");
+  }
+  a.HO = HOc; a.WO = WOc;
+
+  TORCH_CHECK(ktaps == 2, "expected 2 taps per axis for k4s2");
+  dispatch_tile<2, 1, true>(in, w, bias, out, a, st * st);
+  return out;
 }

@@ -26,20 +26,24 @@ std::vector<torch::Tensor> gaussian_kl_bwd(torch::Tensor mu1, torch::Tensor lv1,
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
                               c10::optional<torch::Tensor> bias, long stride,
-                              long pad, long act);
-void conv2d_nhwc_fwd_scatter(torch::Tensor in, torch::Tensor w,
-                             c10::optional<torch::Tensor> bias,
-                             torch::Tensor out, long ipad_h, long ipad_w,
-                             long oys, long oy0, long ox0, long act);
+                              long pad, long act,
+                              c10::optional<torch::Tensor> stats);
+torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
+                                     c10::optional<torch::Tensor> bias,
+                                     long up_stride, long up_pad, long OH,
+                                     long OW, long act);
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd,
-        "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA)");
-  m.def("conv2d_nhwc_fwd_scatter", &conv2d_nhwc_fwd_scatter,
-        "scatter-output stride-1 conv (ConvTranspose fwd / s2 dgrad parity)");
+        "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA)",
+        pybind11::arg("in"), pybind11::arg("w"), pybind11::arg("bias"),
+        pybind11::arg("stride"), pybind11::arg("pad"), pybind11::arg("act"),
+        pybind11::arg("stats") = c10::nullopt);
+  m.def("conv2d_nhwc_fracstride", &conv2d_nhwc_fracstride,
+        "fractionally-strided conv, in-kernel parity loop (ConvT fwd / s2 dgrad)");
   m.def("conv2d_nhwc_wgrad", &conv2d_nhwc_wgrad,
         "NHWC wgrad, split-K over pixels, fp32 workspace (gfx950 MFMA)");
   m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (gfx950)");
