@@ -307,13 +307,16 @@ void dispatch_tile(const torch::Tensor& in, const torch::Tensor& w,
   const bool rsclin = !FRAC && a.C < 32 && KSIZE > 1;
   const int M = a.Nb * a.HO * a.WO;
   const bool small = (long)ceil_div(M, 128) * ceil_div(a.K, 128) < 160;
+  const bool narrow = a.K <= 64;  // half a BN=128 tile would be masked out
   if (rsclin) {
-    if (small && a.K <= 64)
-      launch_one<64, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+    if (narrow)
+      launch_one<128, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
     else
       launch_one<128, 128, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
   } else if (small) {
     launch_one<64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+  } else if (narrow) {
+    launch_one<128, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   } else {
     launch_one<128, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   }

@@ -23,12 +23,13 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
 
-constexpr int BT = 64;     // channel tile (both B and A side)
 constexpr int PCH = 64;    // pixels per staging chunk
 constexpr int THREADS = 256;
 constexpr int PIT = PCH + 8;          // LDS row pitch in elements (16B-aligned)
 constexpr int ROWB = PIT * 2;         // bytes per LDS row
 
+// BT: channel tile on both sides (64 or 128); wave tile BT/2, BT/32 frags.
+template <int BT>
 __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     const __bf16* __restrict__ Y,  // (N, HO, WO, B)
     const __bf16* __restrict__ X,  // (N, H, W, A)
@@ -54,14 +55,15 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
-  const int wm = (wid >> 1) * 32;   // wave row (b) base
-  const int wn = (wid & 1) * 32;    // wave col (a) base
+  constexpr int FRG = BT / 32;
+  const int wm = (wid >> 1) * (BT / 2);   // wave row (b) base
+  const int wn = (wid & 1) * (BT / 2);    // wave col (a) base
 
-  f32x4 acc[2][2];
+  f32x4 acc[FRG][FRG];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < FRG; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FRG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   for (int p0 = p_begin; p0 < p_end; p0 += PCH) {
     // pixel meta for this chunk
@@ -83,12 +85,12 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     }
     __syncthreads();
 
-    // stage Y^T: 64 pix x 8 chunks of 8 channels
+    // stage Y^T: PCH pix x BT/8 chunks of 8 channels
 #pragma unroll
-    for (int it = 0; it < (PCH * 8) / THREADS; ++it) {
+    for (int it = 0; it < (PCH * BT / 8) / THREADS; ++it) {
       const int slot = it * THREADS + tid;
-      const int pix_l = slot >> 3;
-      const int ch0 = (slot & 7) * 8;
+      const int pix_l = slot / (BT / 8);
+      const int ch0 = (slot % (BT / 8)) * 8;
       const int pix = p0 + pix_l;
       bf16x8 v = {};
       if (pix < p_end && b0 + ch0 < B) {
@@ -107,10 +109,10 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     }
     // stage X^T (patch)
 #pragma unroll
-    for (int it = 0; it < (PCH * 8) / THREADS; ++it) {
+    for (int it = 0; it < (PCH * BT / 8) / THREADS; ++it) {
       const int slot = it * THREADS + tid;
-      const int pix_l = slot >> 3;
-      const int ch0 = (slot & 7) * 8;
+      const int pix_l = slot / (BT / 8);
+      const int ch0 = (slot % (BT / 8)) * 8;
       const int hi = pix_hi[pix_l];
       const int wi = pix_wi[pix_l];
       bf16x8 v = {};
@@ -135,18 +137,18 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int pb = (kk * 32 + (lane >> 4) * 8) * 2;  // byte offset along pix
-      bf16x8 a_frag[2], b_frag[2];
+      bf16x8 a_frag[FRG], b_frag[FRG];
 #pragma unroll
-      for (int f = 0; f < 2; ++f) {
+      for (int f = 0; f < FRG; ++f) {
         const int brow = wm + f * 16 + (lane & 15);
         a_frag[f] = *reinterpret_cast<const bf16x8*>(yt + brow * ROWB + pb);
         const int arow = wn + f * 16 + (lane & 15);
         b_frag[f] = *reinterpret_cast<const bf16x8*>(xt + arow * ROWB + pb);
       }
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
+      for (int i = 0; i < FRG; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
+        for (int j = 0; j < FRG; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
     }
@@ -155,11 +157,11 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 
   // epilogue: atomic accumulate into the fp32 workspace
 #pragma unroll
-  for (int j = 0; j < 2; ++j) {
+  for (int j = 0; j < FRG; ++j) {
     const int a = a0 + wn + j * 16 + (lane & 15);
     if (a >= A) continue;
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < FRG; ++i) {
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
         const int b = b0 + wm + i * 16 + (lane >> 4) * 4 + v;
@@ -190,7 +192,8 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
   auto ws = torch::zeros({B, (long)R, (long)S, A},
                          Y.options().dtype(torch::kFloat32));
 
-  const int bt = ceil_div(B, BT), at = ceil_div(A, BT);
+  const int BTsel = (B >= 128 && A >= 128) ? 128 : 64;
+  const int bt = ceil_div(B, BTsel), at = ceil_div(A, BTsel);
   const int p_total = Nb * HO * WO;
   int sp = (int)splitp;
   if (sp <= 0) {
@@ -203,10 +206,18 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
 
   dim3 grid(bt * at, (int)(R * S), sp);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(conv2d_wgrad_kernel, grid, dim3(THREADS), 0, stream,
-                     reinterpret_cast<const __bf16*>(Y.data_ptr()),
-                     reinterpret_cast<const __bf16*>(X.data_ptr()),
-                     ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
-                     (int)S, (int)stride, (int)pad, p_per_slab);
+  if (BTsel == 128) {
+    hipLaunchKernelGGL(conv2d_wgrad_kernel<128>, grid, dim3(THREADS), 0, stream,
+                       reinterpret_cast<const __bf16*>(Y.data_ptr()),
+                       reinterpret_cast<const __bf16*>(X.data_ptr()),
+                       ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
+                       (int)S, (int)stride, (int)pad, p_per_slab);
+  } else {
+    hipLaunchKernelGGL(conv2d_wgrad_kernel<64>, grid, dim3(THREADS), 0, stream,
+                       reinterpret_cast<const __bf16*>(Y.data_ptr()),
+                       reinterpret_cast<const __bf16*>(X.data_ptr()),
+                       ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
+                       (int)S, (int)stride, (int)pad, p_per_slab);
+  }
   return ws;
 }
