@@ -135,39 +135,92 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
   // chunk loop bounds
   const int n_outer = RSCLIN ? 1 : KSIZE * KSIZE;
   const int n_inner = RSCLIN ? (RSC + BK - 1) / BK : (a.C + BK - 1) / BK;
+  const int nchunks = n_outer * n_inner;
 
-  for (int oidx = 0; oidx < n_outer; ++oidx) {
+  constexpr int ASL = (BM * 8) / THREADS;   // A staging slots per thread
+  constexpr int BSL = (BN * 8) / THREADS;   // B staging slots per thread
+  bf16x8 areg[ASL], breg[BSL];
+
+  // T14 software pipeline (guide §6 G15): issue chunk t+1's global loads,
+  // compute chunk t from LDS, then after the barrier write t+1's registers
+  // and immediately issue t+2 — HBM latency hides under the MFMA phase with
+  // a single LDS buffer (34 KB -> 4 blocks/CU occupancy preserved).
+  auto load_chunk = [&](int t) {
+    const int oidx = t / n_inner;
+    const int ci = t - oidx * n_inner;
+    const int c0 = ci * BK;
     const int ro = oidx / KSIZE, so = oidx % KSIZE;
-    for (int ci = 0; ci < n_inner; ++ci) {
-      const int c0 = ci * BK;
-      // ---- stage A ----
 #pragma unroll
-      for (int it = 0; it < (BM * 8) / THREADS; ++it) {
-        const int slot = it * THREADS + tid;
-        const int row = slot >> 3;
-        const int cb = (slot & 7) * 16;
-        bf16x8 v = {};
+    for (int it = 0; it < ASL; ++it) {
+      const int slot = it * THREADS + tid;
+      const int row = slot >> 3;
+      const int cb = (slot & 7) * 16;
+      bf16x8 v = {};
+      if (RSCLIN) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int q = c0 + (cb >> 1) + j;
+          if (q < RSC) {
+            const int rs = q / a.C;
+            const int c = q - rs * a.C;
+            const int r = rs / KSIZE, sx = rs % KSIZE;
+            const int hi = pix_hi[row] + r;
+            const int wi = pix_wi[row] + sx;
+            if (hi >= 0 && hi < a.H && wi >= 0 && wi < a.W)
+              v[j] = in[(long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c];
+          }
+        }
+      } else {
+        const int c = c0 + (cb >> 1);
+        const int hi = pix_hi[row] + ro;
+        const int wi = pix_wi[row] + so;
+        if (hi >= 0 && hi < a.H && wi >= 0 && wi < a.W && c < a.C) {
+          const __bf16* src =
+              in + (long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c;
+          if (cvec && c + 8 <= a.C) {
+            v = *reinterpret_cast<const bf16x8*>(src);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              if (c + j < a.C) v[j] = src[j];
+          }
+        }
+      }
+      areg[it] = v;
+    }
+#pragma unroll
+    for (int it = 0; it < BSL; ++it) {
+      const int slot = it * THREADS + tid;
+      const int row = slot >> 3;
+      const int cb = (slot & 7) * 16;
+      const int k = k0 + row;
+      bf16x8 v = {};
+      if (k < a.K) {
         if (RSCLIN) {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             const int q = c0 + (cb >> 1) + j;
             if (q < RSC) {
-              const int rs = q / a.C;
-              const int c = q - rs * a.C;
-              const int r = rs / KSIZE, s = rs % KSIZE;
-              const int hi = pix_hi[row] + r;
-              const int wi = pix_wi[row] + s;
-              if (hi >= 0 && hi < a.H && wi >= 0 && wi < a.W)
-                v[j] = in[(long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c];
+              long off;
+              if (FRAC) {
+                const int rs = q / a.C;
+                const int c = q - rs * a.C;
+                const int r = a.rmap[par][rs / KSIZE];
+                const int sx = a.smap[par][rs % KSIZE];
+                off = (((long)k * a.wk + r) * a.wk + sx) * a.C + c;
+              } else {
+                off = (long)k * RSC + q;
+              }
+              v[j] = w[off];
             }
           }
         } else {
           const int c = c0 + (cb >> 1);
-          const int hi = pix_hi[row] + ro;
-          const int wi = pix_wi[row] + so;
-          if (hi >= 0 && hi < a.H && wi >= 0 && wi < a.W && c < a.C) {
+          const int r = FRAC ? a.rmap[par][ro] : ro;
+          const int sx = FRAC ? a.smap[par][so] : so;
+          if (c < a.C) {
             const __bf16* src =
-                in + (long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c;
+                w + (((long)k * a.wk + r) * a.wk + sx) * a.C + c;
             if (cvec && c + 8 <= a.C) {
               v = *reinterpret_cast<const bf16x8*>(src);
             } else {
@@ -177,80 +230,60 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
             }
           }
         }
-        *reinterpret_cast<bf16x8*>(a_lds + swz(row, cb)) = v;
       }
-      // ---- stage B (weights) ----
-#pragma unroll
-      for (int it = 0; it < (BN * 8) / THREADS; ++it) {
-        const int slot = it * THREADS + tid;
-        const int row = slot >> 3;
-        const int cb = (slot & 7) * 16;
-        const int k = k0 + row;
-        bf16x8 v = {};
-        if (k < a.K) {
-          if (RSCLIN) {
-            // weight rows are rsc-contiguous; remap taps only when FRAC
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              const int q = c0 + (cb >> 1) + j;
-              if (q < RSC) {
-                int off;
-                if (FRAC) {
-                  const int rs = q / a.C;
-                  const int c = q - rs * a.C;
-                  const int r = a.rmap[par][rs / KSIZE];
-                  const int s = a.smap[par][rs % KSIZE];
-                  off = ((k * a.wk + r) * a.wk + s) * a.C + c;
-                } else {
-                  off = k * RSC + q;
-                }
-                v[j] = w[(long)off];
-              }
-            }
-          } else {
-            const int c = c0 + (cb >> 1);
-            const int r = FRAC ? a.rmap[par][ro] : ro;
-            const int s = FRAC ? a.smap[par][so] : so;
-            if (c < a.C) {
-              const __bf16* src =
-                  w + (((long)k * a.wk + r) * a.wk + s) * a.C + c;
-              if (cvec && c + 8 <= a.C) {
-                v = *reinterpret_cast<const bf16x8*>(src);
-              } else {
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                  if (c + j < a.C) v[j] = src[j];
-              }
-            }
-          }
-        }
-        *reinterpret_cast<bf16x8*>(b_lds + swz(row, cb)) = v;
-      }
-      __syncthreads();
+      breg[it] = v;
+    }
+  };
 
-      // ---- MFMA over the 64-wide chunk (2 x K=32 steps) ----
+  auto write_chunk = [&]() {
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
-        const int cb = kk * 64 + ((lane >> 4) * 16);
-        bf16x8 a_frag[FM], b_frag[FN];
+    for (int it = 0; it < ASL; ++it) {
+      const int slot = it * THREADS + tid;
+      *reinterpret_cast<bf16x8*>(a_lds + swz(slot >> 3, (slot & 7) * 16)) =
+          areg[it];
+    }
 #pragma unroll
-        for (int f = 0; f < FM; ++f) {
-          const int arow = wm + f * 16 + (lane & 15);
-          a_frag[f] = *reinterpret_cast<const bf16x8*>(a_lds + swz(arow, cb));
-        }
+    for (int it = 0; it < BSL; ++it) {
+      const int slot = it * THREADS + tid;
+      *reinterpret_cast<bf16x8*>(b_lds + swz(slot >> 3, (slot & 7) * 16)) =
+          breg[it];
+    }
+  };
+
+  // prologue: chunk 0 into LDS, chunk 1 in flight
+  load_chunk(0);
+  write_chunk();
+  if (nchunks > 1) load_chunk(1);
+  __syncthreads();
+
+  for (int t = 0; t < nchunks; ++t) {
+    // ---- MFMA over the 64-wide chunk (2 x K=32 steps) ----
 #pragma unroll
-        for (int f = 0; f < FN; ++f) {
-          const int brow = wn + f * 16 + (lane & 15);
-          b_frag[f] = *reinterpret_cast<const bf16x8*>(b_lds + swz(brow, cb));
-        }
+    for (int kk = 0; kk < 2; ++kk) {
+      const int cb = kk * 64 + ((lane >> 4) * 16);
+      bf16x8 a_frag[FM], b_frag[FN];
 #pragma unroll
-        for (int i = 0; i < FM; ++i)
-#pragma unroll
-          for (int j = 0; j < FN; ++j)
-            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+      for (int f = 0; f < FM; ++f) {
+        const int arow = wm + f * 16 + (lane & 15);
+        a_frag[f] = *reinterpret_cast<const bf16x8*>(a_lds + swz(arow, cb));
       }
-      __syncthreads();
+#pragma unroll
+      for (int f = 0; f < FN; ++f) {
+        const int brow = wn + f * 16 + (lane & 15);
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(b_lds + swz(brow, cb));
+      }
+#pragma unroll
+      for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    }
+    if (t + 1 < nchunks) {
+      __syncthreads();          // everyone done reading chunk t
+      write_chunk();            // chunk t+1 registers -> LDS
+      if (t + 2 < nchunks) load_chunk(t + 2);  // re-issue immediately
+      __syncthreads();          // chunk t+1 visible
     }
   }
 

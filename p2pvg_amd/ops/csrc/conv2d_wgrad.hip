@@ -36,12 +36,9 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     float* __restrict__ ws,        // (B, R, S, A) fp32, pre-zeroed
     int Nb, int HO, int WO, int B, int H, int W, int A, int R, int S,
     int STRIDE, int PAD, int p_per_slab) {
-  __shared__ __align__(16) char lds[2 * BT * ROWB + PCH * 12];
+  __shared__ __align__(16) char lds[2 * BT * ROWB];
   char* yt = lds;                       // [BT rows (b)][PIT pix]
   char* xt = lds + BT * ROWB;           // [BT rows (a)][PIT pix]
-  int* pix_off = reinterpret_cast<int*>(lds + 2 * BT * ROWB);  // X base offset
-  short* pix_hi = reinterpret_cast<short*>(pix_off + PCH);
-  short* pix_wi = pix_hi + PCH;
 
   const int at_blocks = (A + BT - 1) / BT;
   const int b0 = (blockIdx.x / at_blocks) * BT;
@@ -65,78 +62,83 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 #pragma unroll
     for (int j = 0; j < FRG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  for (int p0 = p_begin; p0 < p_end; p0 += PCH) {
-    // pixel meta for this chunk
-    if (tid < PCH) {
-      const int pix = p0 + tid;
+  constexpr int SL = (PCH * BT / 8) / THREADS;  // staging slots per thread
+  bf16x8 yreg[SL], xreg[SL];
+
+  // T14 pipeline: issue chunk t+1's global loads, MFMA chunk t from LDS,
+  // write t+1 after the barrier. Pixel meta is computed inline per staging
+  // slot (few int divides, hidden under the loads).
+  auto load_chunk = [&](int p0) {
+#pragma unroll
+    for (int it = 0; it < SL; ++it) {
+      const int slot = it * THREADS + tid;
+      const int pix_l = slot / (BT / 8);
+      const int ch0 = (slot % (BT / 8)) * 8;
+      const int pix = p0 + pix_l;
+      bf16x8 vy = {}, vx = {};
       if (pix < p_end) {
         const int n = pix / (HO * WO);
         const int rem = pix - n * (HO * WO);
         const int ho = rem / WO;
         const int wo = rem - ho * WO;
-        pix_off[tid] = n * H * W * A;
-        pix_hi[tid] = (short)(ho * STRIDE - PAD + r);
-        pix_wi[tid] = (short)(wo * STRIDE - PAD + s);
-      } else {
-        pix_off[tid] = 0;
-        pix_hi[tid] = (short)-30000;
-        pix_wi[tid] = (short)-30000;
-      }
-    }
-    __syncthreads();
-
-    // stage Y^T: PCH pix x BT/8 chunks of 8 channels
+        if (b0 + ch0 < B) {
+          const __bf16* src = Y + (long)pix * B + b0 + ch0;
+          if (b0 + ch0 + 8 <= B) {
+            vy = *reinterpret_cast<const bf16x8*>(src);
+          } else {
 #pragma unroll
-    for (int it = 0; it < (PCH * BT / 8) / THREADS; ++it) {
+            for (int j = 0; j < 8; ++j)
+              if (b0 + ch0 + j < B) vy[j] = src[j];
+          }
+        }
+        const int hi = ho * STRIDE - PAD + r;
+        const int wi = wo * STRIDE - PAD + s;
+        if (hi >= 0 && hi < H && wi >= 0 && wi < W && a0 + ch0 < A) {
+          const __bf16* src =
+              X + ((long)n * H * W + (long)hi * W + wi) * A + a0 + ch0;
+          if (a0 + ch0 + 8 <= A) {
+            vx = *reinterpret_cast<const bf16x8*>(src);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+              if (a0 + ch0 + j < A) vx[j] = src[j];
+          }
+        }
+      }
+      yreg[it] = vy;
+      xreg[it] = vx;
+    }
+  };
+
+  auto write_chunk = [&]() {
+    __bf16* yd = reinterpret_cast<__bf16*>(yt);
+    __bf16* xd = reinterpret_cast<__bf16*>(xt);
+#pragma unroll
+    for (int it = 0; it < SL; ++it) {
       const int slot = it * THREADS + tid;
       const int pix_l = slot / (BT / 8);
       const int ch0 = (slot % (BT / 8)) * 8;
-      const int pix = p0 + pix_l;
-      bf16x8 v = {};
-      if (pix < p_end && b0 + ch0 < B) {
-        const __bf16* src = Y + (long)pix * B + b0 + ch0;
-        if (b0 + ch0 + 8 <= B) {
-          v = *reinterpret_cast<const bf16x8*>(src);
-        } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (b0 + ch0 + j < B) v[j] = src[j];
-        }
+      for (int j = 0; j < 8; ++j) {
+        yd[(ch0 + j) * PIT + pix_l] = yreg[it][j];
+        xd[(ch0 + j) * PIT + pix_l] = xreg[it][j];
       }
-      __bf16* dst = reinterpret_cast<__bf16*>(yt);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[(ch0 + j) * PIT + pix_l] = v[j];
     }
-    // stage X^T (patch)
-#pragma unroll
-    for (int it = 0; it < (PCH * BT / 8) / THREADS; ++it) {
-      const int slot = it * THREADS + tid;
-      const int pix_l = slot / (BT / 8);
-      const int ch0 = (slot % (BT / 8)) * 8;
-      const int hi = pix_hi[pix_l];
-      const int wi = pix_wi[pix_l];
-      bf16x8 v = {};
-      if (hi >= 0 && hi < H && wi >= 0 && wi < W && a0 + ch0 < A) {
-        const __bf16* src =
-            X + (long)pix_off[pix_l] + ((long)hi * W + wi) * A + a0 + ch0;
-        if (a0 + ch0 + 8 <= A) {
-          v = *reinterpret_cast<const bf16x8*>(src);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (a0 + ch0 + j < A) v[j] = src[j];
-        }
-      }
-      __bf16* dst = reinterpret_cast<__bf16*>(xt);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dst[(ch0 + j) * PIT + pix_l] = v[j];
-    }
-    __syncthreads();
+  };
 
+  const int nchunks = (p_end - p_begin + PCH - 1) / PCH;
+  if (nchunks > 0) {
+    load_chunk(p_begin);
+    write_chunk();
+    if (nchunks > 1) load_chunk(p_begin + PCH);
+    __syncthreads();
+  }
+
+  for (int t = 0; t < nchunks; ++t) {
     // MFMA: 2 K-steps of 32 pixels
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      const int pb = (kk * 32 + (lane >> 4) * 8) * 2;  // byte offset along pix
+      const int pb = (kk * 32 + (lane >> 4) * 8) * 2;
       bf16x8 a_frag[FRG], b_frag[FRG];
 #pragma unroll
       for (int f = 0; f < FRG; ++f) {
@@ -152,7 +154,12 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    if (t + 1 < nchunks) {
+      __syncthreads();
+      write_chunk();
+      if (t + 2 < nchunks) load_chunk(p_begin + (t + 2) * PCH);
+      __syncthreads();
+    }
   }
 
   // epilogue: atomic accumulate into the fp32 workspace
