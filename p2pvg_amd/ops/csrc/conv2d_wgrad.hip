@@ -28,6 +28,14 @@ constexpr int THREADS = 256;
 constexpr int PIT = PCH + 8;          // LDS row pitch in elements (16B-aligned)
 constexpr int ROWB = PIT * 2;         // bytes per LDS row
 
+// pixel-position XOR swizzle: without it, the transposed staging writes of
+// 16 lanes land on one bank (channel stride 8*PIT*2 B = 0 mod 32 banks,
+// 16-way conflict). XORing the pixel slot by bits of the channel spreads
+// them; reads use the same XOR on their 8-aligned pixel base (guide T2).
+__device__ __forceinline__ int swzp(int ch, int pix) {
+  return pix ^ (((ch >> 3) & 7) << 3);
+}
+
 // BT: channel tile on both sides (64 or 128); wave tile BT/2, BT/32 frags.
 template <int BT>
 __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
@@ -120,8 +128,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
       const int ch0 = (slot % (BT / 8)) * 8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        yd[(ch0 + j) * PIT + pix_l] = yreg[it][j];
-        xd[(ch0 + j) * PIT + pix_l] = xreg[it][j];
+        yd[(ch0 + j) * PIT + swzp(ch0 + j, pix_l)] = yreg[it][j];
+        xd[(ch0 + j) * PIT + swzp(ch0 + j, pix_l)] = xreg[it][j];
       }
     }
   };
@@ -138,14 +146,16 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     // MFMA: 2 K-steps of 32 pixels
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      const int pb = (kk * 32 + (lane >> 4) * 8) * 2;
+      const int pe = kk * 32 + (lane >> 4) * 8;   // pixel base (8-aligned)
       bf16x8 a_frag[FRG], b_frag[FRG];
 #pragma unroll
       for (int f = 0; f < FRG; ++f) {
         const int brow = wm + f * 16 + (lane & 15);
-        a_frag[f] = *reinterpret_cast<const bf16x8*>(yt + brow * ROWB + pb);
+        a_frag[f] = *reinterpret_cast<const bf16x8*>(
+            yt + brow * ROWB + swzp(brow, pe) * 2);
         const int arow = wn + f * 16 + (lane & 15);
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(xt + arow * ROWB + pb);
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(
+            xt + arow * ROWB + swzp(arow, pe) * 2);
       }
 #pragma unroll
       for (int i = 0; i < FRG; ++i)

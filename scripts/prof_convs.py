@@ -100,7 +100,12 @@ def bench_conv(name, N, C, H, W, K, ks, st, pad, transposed=False):
 
     t_ref = timeit(ref_fb)
     t_hip = timeit(hip_fb)
-    print(f"{name:10s} f+b  hip {t_hip:8.1f}us  miopen {t_ref:8.1f}us  x{t_ref/t_hip:5.2f}")
+    # isolated wgrad kernel time
+    if transposed:
+        t_wg = timeit(lambda: ext.conv2d_nhwc_wgrad(xl, g.bfloat16().contiguous(memory_format=CL), ks, ks, st, pad, 0))
+    else:
+        t_wg = timeit(lambda: ext.conv2d_nhwc_wgrad(g.bfloat16().contiguous(memory_format=CL), xl, ks, ks, st, pad, 0))
+    print(f"{name:10s} f+b  hip {t_hip:8.1f}us  miopen {t_ref:8.1f}us  x{t_ref/t_hip:5.2f}  (wgrad {t_wg:7.1f}us)")
 
 
 for row in CONVS:
