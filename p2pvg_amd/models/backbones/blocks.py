@@ -14,6 +14,7 @@ import torch.nn as nn
 
 from ...ops.conv import Conv2d, ConvTranspose2d
 from ...ops.norm import BatchNorm2d
+from ...ops.fused_norm import FusedSequential
 
 
 class dcgan_conv(nn.Module):
@@ -21,7 +22,7 @@ class dcgan_conv(nn.Module):
 
     def __init__(self, nin: int, nout: int):
         super().__init__()
-        self.main = nn.Sequential(
+        self.main = FusedSequential(
             Conv2d(nin, nout, 4, 2, 1),
             BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
@@ -36,7 +37,7 @@ class dcgan_upconv(nn.Module):
 
     def __init__(self, nin: int, nout: int):
         super().__init__()
-        self.main = nn.Sequential(
+        self.main = FusedSequential(
             ConvTranspose2d(nin, nout, 4, 2, 1),
             BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
@@ -51,7 +52,7 @@ class vgg_layer(nn.Module):
 
     def __init__(self, nin: int, nout: int):
         super().__init__()
-        self.main = nn.Sequential(
+        self.main = FusedSequential(
             Conv2d(nin, nout, 3, 1, 1),
             BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
@@ -67,8 +68,8 @@ class residual_linear(nn.Module):
 
     def __init__(self, nin: int, nout: int):
         super().__init__()
-        self.shortcut = nn.Sequential(nn.Linear(nin, nout), nn.ReLU(inplace=True))
-        self.long_path = nn.Sequential(
+        self.shortcut = FusedSequential(nn.Linear(nin, nout), nn.ReLU(inplace=True))
+        self.long_path = FusedSequential(
             nn.Linear(nin, nin // 2),
             nn.ReLU(inplace=True),
             nn.Linear(nin // 2, nin // 2),

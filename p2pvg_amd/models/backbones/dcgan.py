@@ -12,6 +12,7 @@ import torch.nn as nn
 
 from ...ops.conv import Conv2d, ConvTranspose2d
 from ...ops.norm import BatchNorm2d
+from ...ops.fused_norm import FusedSequential
 
 from .blocks import dcgan_conv, dcgan_upconv
 
@@ -25,7 +26,7 @@ class Encoder64(nn.Module):
         self.c2 = dcgan_conv(nf, nf * 2)          # -> 128x16x16
         self.c3 = dcgan_conv(nf * 2, nf * 4)      # -> 256x8x8
         self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x4x4
-        self.c5 = nn.Sequential(                  # -> dim x1x1
+        self.c5 = FusedSequential(                  # -> dim x1x1
             Conv2d(nf * 8, dim, 4, 1, 0),
             BatchNorm2d(dim),
             nn.Tanh(),
@@ -45,7 +46,7 @@ class Decoder64(nn.Module):
         super().__init__()
         self.dim = dim
         nf = 64
-        self.upc1 = nn.Sequential(
+        self.upc1 = FusedSequential(
             ConvTranspose2d(dim, nf * 8, 4, 1, 0),
             BatchNorm2d(nf * 8),
             nn.LeakyReLU(0.2, inplace=True),
@@ -53,7 +54,7 @@ class Decoder64(nn.Module):
         self.upc2 = dcgan_upconv(nf * 8 * 2, nf * 4)
         self.upc3 = dcgan_upconv(nf * 4 * 2, nf * 2)
         self.upc4 = dcgan_upconv(nf * 2 * 2, nf)
-        self.upc5 = nn.Sequential(
+        self.upc5 = FusedSequential(
             ConvTranspose2d(nf * 2, nc, 4, 2, 1),
             nn.Sigmoid(),
         )
@@ -79,7 +80,7 @@ class Encoder128(nn.Module):
         self.c3 = dcgan_conv(nf * 2, nf * 4)      # -> 256x16
         self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x8
         self.c5 = dcgan_conv(nf * 8, nf * 8)      # -> 512x4
-        self.c6 = nn.Sequential(
+        self.c6 = FusedSequential(
             Conv2d(nf * 8, dim, 4, 1, 0),
             BatchNorm2d(dim),
             nn.Tanh(),
@@ -100,7 +101,7 @@ class Decoder128(nn.Module):
         super().__init__()
         self.dim = dim
         nf = 64
-        self.upc1 = nn.Sequential(
+        self.upc1 = FusedSequential(
             ConvTranspose2d(dim, nf * 8, 4, 1, 0),
             BatchNorm2d(nf * 8),
             nn.LeakyReLU(0.2, inplace=True),
@@ -109,7 +110,7 @@ class Decoder128(nn.Module):
         self.upc3 = dcgan_upconv(nf * 8 * 2, nf * 4)
         self.upc4 = dcgan_upconv(nf * 4 * 2, nf * 2)
         self.upc5 = dcgan_upconv(nf * 2 * 2, nf)
-        self.upc6 = nn.Sequential(
+        self.upc6 = FusedSequential(
             ConvTranspose2d(nf * 2, nc, 4, 2, 1),
             nn.Sigmoid(),
         )

@@ -12,6 +12,7 @@ import torch.nn as nn
 
 from ...ops.conv import Conv2d, ConvTranspose2d
 from ...ops.norm import BatchNorm2d
+from ...ops.fused_norm import FusedSequential
 
 from .blocks import vgg_layer
 
@@ -20,15 +21,15 @@ class Encoder64(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
-        self.c1 = nn.Sequential(vgg_layer(nc, 64), vgg_layer(64, 64))
-        self.c2 = nn.Sequential(vgg_layer(64, 128), vgg_layer(128, 128))
-        self.c3 = nn.Sequential(
+        self.c1 = FusedSequential(vgg_layer(nc, 64), vgg_layer(64, 64))
+        self.c2 = FusedSequential(vgg_layer(64, 128), vgg_layer(128, 128))
+        self.c3 = FusedSequential(
             vgg_layer(128, 256), vgg_layer(256, 256), vgg_layer(256, 256)
         )
-        self.c4 = nn.Sequential(
+        self.c4 = FusedSequential(
             vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
-        self.c5 = nn.Sequential(
+        self.c5 = FusedSequential(
             Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
         )
         self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
@@ -46,19 +47,19 @@ class Decoder64(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
-        self.upc1 = nn.Sequential(
+        self.upc1 = FusedSequential(
             ConvTranspose2d(dim, 512, 4, 1, 0),
             BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
-        self.upc2 = nn.Sequential(
+        self.upc2 = FusedSequential(
             vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 256)
         )
-        self.upc3 = nn.Sequential(
+        self.upc3 = FusedSequential(
             vgg_layer(256 * 2, 256), vgg_layer(256, 256), vgg_layer(256, 128)
         )
-        self.upc4 = nn.Sequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
-        self.upc5 = nn.Sequential(
+        self.upc4 = FusedSequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
+        self.upc5 = FusedSequential(
             vgg_layer(64 * 2, 64),
             ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
@@ -78,18 +79,18 @@ class Encoder128(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
-        self.c1 = nn.Sequential(vgg_layer(nc, 64), vgg_layer(64, 64))
-        self.c2 = nn.Sequential(vgg_layer(64, 128), vgg_layer(128, 128))
-        self.c3 = nn.Sequential(
+        self.c1 = FusedSequential(vgg_layer(nc, 64), vgg_layer(64, 64))
+        self.c2 = FusedSequential(vgg_layer(64, 128), vgg_layer(128, 128))
+        self.c3 = FusedSequential(
             vgg_layer(128, 256), vgg_layer(256, 256), vgg_layer(256, 256)
         )
-        self.c4 = nn.Sequential(
+        self.c4 = FusedSequential(
             vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
-        self.c5 = nn.Sequential(
+        self.c5 = FusedSequential(
             vgg_layer(512, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
-        self.c6 = nn.Sequential(
+        self.c6 = FusedSequential(
             Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
         )
         self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
@@ -108,22 +109,22 @@ class Decoder128(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
-        self.upc1 = nn.Sequential(
+        self.upc1 = FusedSequential(
             ConvTranspose2d(dim, 512, 4, 1, 0),
             BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
-        self.upc2 = nn.Sequential(
+        self.upc2 = FusedSequential(
             vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 512)
         )
-        self.upc3 = nn.Sequential(
+        self.upc3 = FusedSequential(
             vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 256)
         )
-        self.upc4 = nn.Sequential(
+        self.upc4 = FusedSequential(
             vgg_layer(256 * 2, 256), vgg_layer(256, 256), vgg_layer(256, 128)
         )
-        self.upc5 = nn.Sequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
-        self.upc6 = nn.Sequential(
+        self.upc5 = FusedSequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
+        self.upc6 = FusedSequential(
             vgg_layer(64 * 2, 64),
             ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),

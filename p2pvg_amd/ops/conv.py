@@ -45,38 +45,80 @@ def _nhwc_flat(t: torch.Tensor) -> torch.Tensor:
     return t.permute(0, 2, 3, 1).reshape(n, h * w * c)
 
 
+def _act_fwd_torch(out, act):
+    if act == 1:
+        return torch.where(out > 0, out, out * 0.2)
+    if act == 2:
+        return torch.tanh(out)
+    if act == 3:
+        return torch.sigmoid(out)
+    return out
+
+
+def _act_bwd_from_y(dy, y, act):
+    """grad through the fused epilogue activation, from the post-act value."""
+    if act == 1:
+        return torch.where(y > 0, dy, dy * 0.2)
+    if act == 2:
+        return dy * (1 - y.float() * y.float()).to(dy.dtype)
+    if act == 3:
+        yf = y.float()
+        return dy * (yf * (1 - yf)).to(dy.dtype)
+    return dy
+
+
+def _gemm_stats(out2):
+    f = out2.float()
+    return torch.stack([f.sum(0), (f * f).sum(0)]).contiguous()
+
+
 class Conv2dNHWCFn(torch.autograd.Function):
-    """y = conv2d(x, w, b, stride, pad) with x, w bf16 channels_last."""
+    """y = act(conv2d(x, w, b, stride, pad)); optional per-channel sum/sumsq
+    stats of the output (for the fused BatchNorm). Returns (y, stats)."""
 
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, pad: int):
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int = 0,
+                want_stats: bool = False):
         ext = _ext()
         k = w.shape[2]
         b32 = b.float() if b is not None else None
+        stats = (
+            torch.zeros(2, w.shape[0], device=x.device, dtype=torch.float32)
+            if want_stats else None
+        )
         # degenerate whole-image conv (k == H, pad 0): plain GEMM
         gemm = pad == 0 and k == x.shape[2] and k == x.shape[3]
         if gemm:
             out = torch.mm(_nhwc_flat(x), _nhwc_flat(w).t())
             if b is not None:
                 out = out + b.to(out.dtype)
+            out = _act_fwd_torch(out, act)
+            if want_stats:
+                stats = _gemm_stats(out)
             out = out.view(x.shape[0], w.shape[0], 1, 1).contiguous(
                 memory_format=CL
             )
         else:
-            out = ext.conv2d_nhwc_fwd(x, w, b32, stride, pad, 0, None)
-        ctx.save_for_backward(x, w)
+            out = ext.conv2d_nhwc_fwd(x, w, b32, stride, pad, act, stats)
+        ctx.save_for_backward(x, w, out if act != 0 else None)
         ctx.stride, ctx.pad, ctx.has_bias, ctx.gemm = stride, pad, b is not None, gemm
-        return out
+        ctx.act = act
+        if stats is None:
+            stats = torch.empty(0, device=x.device)
+        ctx.mark_non_differentiable(stats)
+        return out, stats
 
     @staticmethod
-    def backward(ctx, gout):
-        x, w = ctx.saved_tensors
+    def backward(ctx, gout, _gstats):
+        x, w, y = ctx.saved_tensors
         stride, pad = ctx.stride, ctx.pad
         k = w.shape[2]
         ext = _ext()
         gout = gout.contiguous(memory_format=CL)
         if gout.dtype != torch.bfloat16:
             gout = gout.to(torch.bfloat16)
+        if ctx.act != 0:
+            gout = _act_bwd_from_y(gout, y, ctx.act).contiguous(memory_format=CL)
 
         dx = dw = db = None
         if ctx.gemm:
@@ -91,7 +133,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
                 ).permute(0, 3, 1, 2)
             if ctx.has_bias and ctx.needs_input_grad[2]:
                 db = g2.float().sum(0)
-            return dx, dw, db, None, None
+            return dx, dw, db, None, None, None, None
 
         if ctx.needs_input_grad[0]:
             if stride == 1:
@@ -100,7 +142,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
             else:
                 wt = w.transpose(0, 1).contiguous(memory_format=CL)  # (C,K,k,k)
                 dx = ext.conv2d_nhwc_fracstride(
-                    gout, wt, None, stride, pad, x.shape[2], x.shape[3], 0
+                    gout, wt, None, stride, pad, x.shape[2], x.shape[3], 0, None
                 )
         if ctx.needs_input_grad[1]:
             ws = ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0)
@@ -108,19 +150,24 @@ class Conv2dNHWCFn(torch.autograd.Function):
                 .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = gout.float().sum(dim=(0, 2, 3))
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, None, None
 
 
 class ConvT2dNHWCFn(torch.autograd.Function):
     """y = conv_transpose2d(x, w, b, stride, pad); w logical (Ci, Co, k, k)."""
 
     @staticmethod
-    def forward(ctx, x, w, b, stride: int, pad: int):
+    def forward(ctx, x, w, b, stride: int, pad: int, act: int = 0,
+                want_stats: bool = False):
         ext = _ext()
         k = w.shape[2]
         co = w.shape[1]
         n, _, h, wdt = x.shape
         b32 = b.float() if b is not None else None
+        stats = (
+            torch.zeros(2, co, device=x.device, dtype=torch.float32)
+            if want_stats else None
+        )
         gemm = stride == 1 and pad == 0 and h == 1 and wdt == 1
         if gemm:
             # 1x1 -> kxk: out[n, y, x, co] = sum_ci in[n,ci] w[ci,co,y,x].
@@ -129,28 +176,38 @@ class ConvT2dNHWCFn(torch.autograd.Function):
             out = torch.mm(_nhwc_flat(x), _nhwc_flat(w))
             if b is not None:
                 out = out.view(n, k * k, co) + b.to(out.dtype)
+            out = _act_fwd_torch(out, act)
+            if want_stats:
+                stats = _gemm_stats(out.reshape(n * k * k, co))
             out = out.view(n, k, k, co).permute(0, 3, 1, 2)
         elif stride == 1:
             wt = w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)  # (Co,Ci,k,k)
-            out = ext.conv2d_nhwc_fwd(x, wt, b32, 1, k - 1 - pad, 0, None)
+            out = ext.conv2d_nhwc_fwd(x, wt, b32, 1, k - 1 - pad, act, stats)
         else:
             oh = (h - 1) * stride - 2 * pad + k
             ow = (wdt - 1) * stride - 2 * pad + k
             wt = w.transpose(0, 1).contiguous(memory_format=CL)  # (Co, Ci, k, k)
-            out = ext.conv2d_nhwc_fracstride(x, wt, b32, stride, pad, oh, ow, 0)
-        ctx.save_for_backward(x, w)
+            out = ext.conv2d_nhwc_fracstride(x, wt, b32, stride, pad, oh, ow,
+                                             act, stats)
+        ctx.save_for_backward(x, w, out if act != 0 else None)
         ctx.stride, ctx.pad, ctx.has_bias, ctx.gemm = stride, pad, b is not None, gemm
-        return out
+        ctx.act = act
+        if stats is None:
+            stats = torch.empty(0, device=x.device)
+        ctx.mark_non_differentiable(stats)
+        return out, stats
 
     @staticmethod
-    def backward(ctx, gout):
-        x, w = ctx.saved_tensors
+    def backward(ctx, gout, _gstats):
+        x, w, y = ctx.saved_tensors
         stride, pad = ctx.stride, ctx.pad
         k = w.shape[2]
         ext = _ext()
         gout = gout.contiguous(memory_format=CL)
         if gout.dtype != torch.bfloat16:
             gout = gout.to(torch.bfloat16)
+        if ctx.act != 0:
+            gout = _act_bwd_from_y(gout, y, ctx.act).contiguous(memory_format=CL)
 
         dx = dw = db = None
         if ctx.gemm:
@@ -165,7 +222,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 dw = dwf.view(x.shape[1], k, k, w.shape[1]).permute(0, 3, 1, 2)
             if ctx.has_bias and ctx.needs_input_grad[2]:
                 db = gout.float().sum(dim=(0, 2, 3))
-            return dx, dw, db, None, None
+            return dx, dw, db, None, None, None, None
 
         if ctx.needs_input_grad[0]:
             # dgrad of convT = plain conv with the untransposed weight
@@ -177,7 +234,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = gout.float().sum(dim=(0, 2, 3))
-        return dx, dw, db, None, None
+        return dx, dw, db, None, None, None, None
 
 
 def _use_hip_path(x: torch.Tensor) -> bool:
@@ -208,9 +265,10 @@ class Conv2d(nn.Conv2d):
             with torch.autocast("cuda", enabled=False):
                 xl = _to_cl_bf16(x)
                 wl = _to_cl_bf16(self.weight)
-                return Conv2dNHWCFn.apply(
-                    xl, wl, self.bias, self.stride[0], self.padding[0]
+                out, _ = Conv2dNHWCFn.apply(
+                    xl, wl, self.bias, self.stride[0], self.padding[0], 0, False
                 )
+                return out
         return super().forward(x)
 
 
@@ -233,7 +291,8 @@ class ConvTranspose2d(nn.ConvTranspose2d):
             with torch.autocast("cuda", enabled=False):
                 xl = _to_cl_bf16(x)
                 wl = _to_cl_bf16(self.weight)
-                return ConvT2dNHWCFn.apply(
-                    xl, wl, self.bias, self.stride[0], self.padding[0]
+                out, _ = ConvT2dNHWCFn.apply(
+                    xl, wl, self.bias, self.stride[0], self.padding[0], 0, False
                 )
+                return out
         return super().forward(x, output_size)

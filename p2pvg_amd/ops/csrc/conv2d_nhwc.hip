@@ -418,7 +418,8 @@ torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
 torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
                                      c10::optional<torch::Tensor> bias,
                                      long up_stride, long up_pad, long OH,
-                                     long OW, long act) {
+                                     long OW, long act,
+                                     c10::optional<torch::Tensor> stats) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
@@ -438,7 +439,7 @@ torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
   a.Nb = Nb; a.H = H; a.W = W; a.C = C; a.K = K;
   a.OH = (int)OH; a.OW = (int)OW;
   a.act = (int)act; a.wk = WK; a.oys = st;
-  a.stats = nullptr;
+  a.stats = stats.has_value() ? stats->data_ptr<float>() : nullptr;
 
   // taps per parity: r with (p + pad - r) % st == 0, offset (p+pad-r)/st
   int ktaps = -1;

@@ -31,7 +31,20 @@ torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
 torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
                                      c10::optional<torch::Tensor> bias,
                                      long up_stride, long up_pad, long OH,
-                                     long OW, long act);
+                                     long OW, long act,
+                                     c10::optional<torch::Tensor> stats);
+std::vector<torch::Tensor> bn_act_fwd_train(
+    torch::Tensor x, torch::Tensor stats, torch::Tensor gamma,
+    torch::Tensor beta, c10::optional<torch::Tensor> running_mean,
+    c10::optional<torch::Tensor> running_var, double momentum, double eps,
+    long act);
+torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
+                              torch::Tensor beta, torch::Tensor running_mean,
+                              torch::Tensor running_var, double eps, long act);
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor y,
+                                      torch::Tensor dy, torch::Tensor mean,
+                                      torch::Tensor invstd, torch::Tensor scale,
+                                      long act);
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp);
 
@@ -43,7 +56,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("stride"), pybind11::arg("pad"), pybind11::arg("act"),
         pybind11::arg("stats") = c10::nullopt);
   m.def("conv2d_nhwc_fracstride", &conv2d_nhwc_fracstride,
-        "fractionally-strided conv, in-kernel parity loop (ConvT fwd / s2 dgrad)");
+        "fractionally-strided conv, in-kernel parity loop (ConvT fwd / s2 dgrad)",
+        pybind11::arg("in"), pybind11::arg("w"), pybind11::arg("bias"),
+        pybind11::arg("up_stride"), pybind11::arg("up_pad"), pybind11::arg("OH"),
+        pybind11::arg("OW"), pybind11::arg("act"),
+        pybind11::arg("stats") = c10::nullopt);
+  m.def("bn_act_fwd_train", &bn_act_fwd_train, "fused BN+act train fwd (gfx950)");
+  m.def("bn_act_fwd_eval", &bn_act_fwd_eval, "fused BN+act eval fwd (gfx950)");
+  m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act bwd (gfx950)");
   m.def("conv2d_nhwc_wgrad", &conv2d_nhwc_wgrad,
         "NHWC wgrad, split-K over pixels, fp32 workspace (gfx950 MFMA)");
   m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (gfx950)");

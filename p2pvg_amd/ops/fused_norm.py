@@ -1,0 +1,167 @@
+"""Fused Conv -> BatchNorm -> activation path (SURVEY §2.6 K1+K4+K5).
+
+The conv kernel's epilogue accumulates per-channel sum/sumsq, so the whole
+block costs: conv (one kernel) + BN finalize (K threads) + one elementwise
+normalize+activation pass. Backward folds the activation grad into the BN
+reduction/apply kernels. `FusedSequential` pattern-matches the reference's
+Sequential(conv, bn, act) blocks (state_dict keys unchanged) and routes them
+here on the gfx950 path.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+CL = torch.channels_last
+
+
+def _ext():
+    from . import _hip_ext_loader
+
+    return _hip_ext_loader.load()
+
+
+ACT_NONE, ACT_LEAKY, ACT_TANH, ACT_SIGMOID = 0, 1, 2, 3
+
+
+def act_code(mod: nn.Module) -> Optional[int]:
+    if isinstance(mod, nn.LeakyReLU) and abs(mod.negative_slope - 0.2) < 1e-8:
+        return ACT_LEAKY
+    if isinstance(mod, nn.Tanh):
+        return ACT_TANH
+    if isinstance(mod, nn.Sigmoid):
+        return ACT_SIGMOID
+    return None
+
+
+class FusedBNActFn(torch.autograd.Function):
+    """y = act(bn(x)) with batch statistics from a pre-accumulated (2,C)
+    sum/sumsq buffer (conv epilogue) and fused activation backward."""
+
+    @staticmethod
+    def forward(ctx, x, stats, gamma, beta, running_mean, running_var,
+                momentum: float, eps: float, act: int):
+        ext = _ext()
+        y, mean, invstd, scale = ext.bn_act_fwd_train(
+            x, stats, gamma.float(), beta.float(), running_mean, running_var,
+            momentum, eps, act,
+        )
+        ctx.save_for_backward(x, y, mean, invstd, scale)
+        ctx.act = act
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, mean, invstd, scale = ctx.saved_tensors
+        ext = _ext()
+        dy = dy.contiguous(memory_format=CL)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        dx, dgamma, dbeta = ext.bn_act_bwd(x, y, dy, mean, invstd, scale, ctx.act)
+        return dx, None, dgamma, dbeta, None, None, None, None, None
+
+
+def fused_conv_bn_act(x, conv, bn, act: int):
+    """conv (with stats epilogue) -> fused BN+act. Training and eval modes."""
+    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _to_cl_bf16
+
+    ext = _ext()
+    training = bn.training
+    with torch.autocast("cuda", enabled=False):
+        xl = _to_cl_bf16(x)
+        wl = _to_cl_bf16(conv.weight)
+        want_stats = training
+        if isinstance(conv, nn.ConvTranspose2d):
+            out, stats = ConvT2dNHWCFn.apply(
+                xl, wl, conv.bias, conv.stride[0], conv.padding[0], 0, want_stats
+            )
+        else:
+            out, stats = Conv2dNHWCFn.apply(
+                xl, wl, conv.bias, conv.stride[0], conv.padding[0], 0, want_stats
+            )
+        if not want_stats:
+            stats = None
+        if training:
+            return FusedBNActFn.apply(
+                out, stats, bn.weight, bn.bias,
+                bn.running_mean if bn.track_running_stats else None,
+                bn.running_var if bn.track_running_stats else None,
+                bn.momentum if bn.momentum is not None else 0.1,
+                bn.eps, act,
+            )
+        return ext.bn_act_fwd_eval(
+            out, bn.weight.float(), bn.bias.float(), bn.running_mean,
+            bn.running_var, bn.eps, act,
+        )
+
+
+def fused_conv_act(x, conv, act: int):
+    """conv with the activation fused straight into the epilogue (no BN)."""
+    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _to_cl_bf16
+
+    with torch.autocast("cuda", enabled=False):
+        xl = _to_cl_bf16(x)
+        wl = _to_cl_bf16(conv.weight)
+        if isinstance(conv, nn.ConvTranspose2d):
+            out, _ = ConvT2dNHWCFn.apply(
+                xl, wl, conv.bias, conv.stride[0], conv.padding[0], act, False
+            )
+        else:
+            out, _ = Conv2dNHWCFn.apply(
+                xl, wl, conv.bias, conv.stride[0], conv.padding[0], act, False
+            )
+        return out
+
+
+def _conv_supported(conv, x) -> bool:
+    if isinstance(conv, nn.ConvTranspose2d):
+        extra = conv.output_padding == (0, 0)
+    else:
+        extra = True
+    return (
+        extra
+        and conv.kernel_size[0] == conv.kernel_size[1]
+        and conv.stride[0] == conv.stride[1]
+        and conv.padding[0] == conv.padding[1]
+        and conv.kernel_size[0] in (1, 2, 3, 4)
+        and conv.stride[0] in (1, 2)
+        and conv.dilation == (1, 1)
+        and conv.groups == 1
+    )
+
+
+class FusedSequential(nn.Sequential):
+    """nn.Sequential that fuses (conv [, batchnorm] [, activation]) chains on
+    the gfx950 path. Same children indices -> same state_dict keys. Greedy
+    scan, so trailing conv+act pairs inside longer chains fuse too."""
+
+    def forward(self, x):
+        from .conv import _use_hip_path
+
+        mods = list(self)
+        i = 0
+        while i < len(mods):
+            m = mods[i]
+            if (
+                isinstance(m, (nn.Conv2d, nn.ConvTranspose2d))
+                and _use_hip_path(x)
+                and _conv_supported(m, x)
+            ):
+                if (
+                    i + 2 < len(mods)
+                    and isinstance(mods[i + 1], nn.BatchNorm2d)
+                    and act_code(mods[i + 2]) is not None
+                    and mods[i + 1].num_features % 8 == 0
+                ):
+                    x = fused_conv_bn_act(x, m, mods[i + 1], act_code(mods[i + 2]))
+                    i += 3
+                    continue
+                if i + 1 < len(mods) and act_code(mods[i + 1]) is not None:
+                    x = fused_conv_act(x, m, act_code(mods[i + 1]))
+                    i += 2
+                    continue
+            x = m(x)
+            i += 1
+        return x

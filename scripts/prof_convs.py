@@ -76,16 +76,16 @@ def bench_conv(name, N, C, H, W, K, ks, st, pad, transposed=False):
 
     # fwd
     t_ref = timeit(ref_f)
-    t_hip = timeit(lambda: Fn.apply(xl, wl, None, st, pad))
+    t_hip = timeit(lambda: Fn.apply(xl, wl, None, st, pad, 0, False)[0])
     print(f"{name:10s} fwd  hip {t_hip:8.1f}us  miopen {t_ref:8.1f}us  x{t_ref/t_hip:5.2f}")
 
     # full fwd+bwd
     xg = xl.detach().requires_grad_()
     wg = wl.detach().requires_grad_()
-    g = torch.randn_like(Fn.apply(xg, wg, None, st, pad))
+    g = torch.randn_like(Fn.apply(xg, wg, None, st, pad, 0, False)[0])
 
     def hip_fb():
-        out = Fn.apply(xg, wg, None, st, pad)
+        out, _ = Fn.apply(xg, wg, None, st, pad, 0, False)
         out.backward(g)
 
     xr = xl.detach().requires_grad_()
