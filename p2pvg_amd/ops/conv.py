@@ -83,7 +83,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
         k = w.shape[2]
         b32 = b.float() if b is not None else None
         stats = (
-            torch.zeros(2, w.shape[0], device=x.device, dtype=torch.float32)
+            torch.zeros(64, 2, w.shape[0], device=x.device, dtype=torch.float32)
             if want_stats else None
         )
         # degenerate whole-image conv (k == H, pad 0): plain GEMM
@@ -94,7 +94,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
                 out = out + b.to(out.dtype)
             out = _act_fwd_torch(out, act)
             if want_stats:
-                stats = _gemm_stats(out)
+                stats = _gemm_stats(out).unsqueeze(0)
             out = out.view(x.shape[0], w.shape[0], 1, 1).contiguous(
                 memory_format=CL
             )
@@ -165,7 +165,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         n, _, h, wdt = x.shape
         b32 = b.float() if b is not None else None
         stats = (
-            torch.zeros(2, co, device=x.device, dtype=torch.float32)
+            torch.zeros(64, 2, co, device=x.device, dtype=torch.float32)
             if want_stats else None
         )
         gemm = stride == 1 and pad == 0 and h == 1 and wdt == 1
@@ -178,7 +178,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 out = out.view(n, k * k, co) + b.to(out.dtype)
             out = _act_fwd_torch(out, act)
             if want_stats:
-                stats = _gemm_stats(out.reshape(n * k * k, co))
+                stats = _gemm_stats(out.reshape(n * k * k, co)).unsqueeze(0)
             out = out.view(n, k, k, co).permute(0, 3, 1, 2)
         elif stride == 1:
             wt = w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)  # (Co,Ci,k,k)

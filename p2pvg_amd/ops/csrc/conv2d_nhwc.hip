@@ -309,8 +309,18 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
       }
     }
     if (a.stats != nullptr) {
-      atomicAdd(&a.stats[col], csum);
-      atomicAdd(&a.stats[a.K + col], csq);
+      // reduce the 4 lanes sharing this column (l, l+16, l+32, l+48), then
+      // one atomic per column into a per-block bucket (64 copies cut the
+      // same-address contention that serializes large-M launches)
+      csum += __shfl_xor(csum, 16, 64);
+      csum += __shfl_xor(csum, 32, 64);
+      csq += __shfl_xor(csq, 16, 64);
+      csq += __shfl_xor(csq, 32, 64);
+      if ((lane >> 4) == 0) {
+        float* bucket = a.stats + (long)(blockIdx.x & 63) * 2 * a.K;
+        atomicAdd(&bucket[col], csum);
+        atomicAdd(&bucket[a.K + col], csq);
+      }
     }
   }
 }

@@ -52,11 +52,16 @@ __global__ void bn_finalize_kernel(const float* __restrict__ stats,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
                                    float momentum, float eps, float count,
-                                   int K) {
+                                   int K, int nbuckets) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= K) return;
-  const float m = stats[c] / count;
-  float var = stats[K + c] / count - m * m;
+  float s0 = 0.f, s1 = 0.f;
+  for (int b = 0; b < nbuckets; ++b) {
+    s0 += stats[(long)b * 2 * K + c];
+    s1 += stats[(long)b * 2 * K + K + c];
+  }
+  const float m = s0 / count;
+  float var = s1 / count - m * m;
   var = var > 0.f ? var : 0.f;
   const float inv = rsqrtf(var + eps);
   mean_out[c] = m;
@@ -111,37 +116,43 @@ __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
   }
 }
 
-// s1[c] = sum dy', s2[c] = sum dy'*xhat with dy' = dy * dact(y)
+// s1[c] = sum dy', s2[c] = sum dy'*xhat with dy' = dy * dact(y).
+// C/8 divides BLOCK and the grid stride, so every thread's channel group is
+// FIXED across its whole grid-stride walk: partial sums live in 16 registers
+// and each thread issues exactly 16 atomics at the end.
 __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ y,
     const __bf16* __restrict__ dy, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ red,  // (2,K)
     long nvec, int C, int act) {
   const int cvec = C / 8;
-  // per-thread partial sums for 8 channels at a time won't work when a
-  // thread visits many channel groups; accumulate via LDS per block.
-  extern __shared__ float sred[];  // 2*C floats
-  for (int i = threadIdx.x; i < 2 * C; i += BLOCK) sred[i] = 0.f;
-  __syncthreads();
+  const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
+  float p1[8], p2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) p1[j] = p2[j] = 0.f;
+  float mn[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mn[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+  }
 
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    const int c0 = (int)(i % cvec) * 8;
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 yv = *reinterpret_cast<const bf16x8*>(y + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c0 + j;
       const float dyp = (float)gv[j] * act_bwd_from_y((float)yv[j], act);
-      const float xhat = ((float)xv[j] - mean[c]) * invstd[c];
-      atomicAdd(&sred[c], dyp);
-      atomicAdd(&sred[C + c], dyp * xhat);
+      p1[j] += dyp;
+      p2[j] += dyp * (((float)xv[j] - mn[j]) * is[j]);
     }
   }
-  __syncthreads();
-  for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-    if (sred[i] != 0.f) atomicAdd(&red[i], sred[i]);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    atomicAdd(&red[c0 + j], p1[j]);
+    atomicAdd(&red[C + c0 + j], p2[j]);
   }
 }
 
@@ -200,6 +211,7 @@ std::vector<torch::Tensor> bn_act_fwd_train(
   auto y = torch::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
 
+  const int nbuckets = stats.dim() == 3 ? (int)stats.size(0) : 1;
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
                      stream, stats.data_ptr<float>(), gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), mean.data_ptr<float>(),
@@ -207,7 +219,7 @@ std::vector<torch::Tensor> bn_act_fwd_train(
                      shift.data_ptr<float>(),
                      running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
                      running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
-                     (float)momentum, (float)eps, (float)count, C);
+                     (float)momentum, (float)eps, (float)count, C, nbuckets);
 
   const long nvec = x.numel() / 8;
   hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK), 0,
@@ -258,7 +270,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor y,
   auto stream = at::cuda::getCurrentCUDAStream();
   const long nvec = x.numel() / 8;
   hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(pick_grid(nvec)),
-                     dim3(BLOCK), 2 * C * sizeof(float), stream,
+                     dim3(BLOCK), 0, stream,
                      reinterpret_cast<const __bf16*>(x.data_ptr()),
                      reinterpret_cast<const __bf16*>(y.data_ptr()),
                      reinterpret_cast<const __bf16*>(dy.data_ptr()),

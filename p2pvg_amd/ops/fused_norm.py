@@ -154,6 +154,7 @@ class FusedSequential(nn.Sequential):
                     and isinstance(mods[i + 1], nn.BatchNorm2d)
                     and act_code(mods[i + 2]) is not None
                     and mods[i + 1].num_features % 8 == 0
+                    and 256 % (mods[i + 1].num_features // 8) == 0
                 ):
                     x = fused_conv_bn_act(x, m, mods[i + 1], act_code(mods[i + 2]))
                     i += 3
