@@ -291,8 +291,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 #pragma unroll
   for (int j = 0; j < FN; ++j) {
     const int col = k0 + wn + j * 16 + (lane & 15);
-    if (col >= a.K) continue;
-    const float bv = bias != nullptr ? bias[col] : 0.f;
+    const bool colv = col < a.K;
+    const float bv = (colv && bias != nullptr) ? bias[col] : 0.f;
     float csum = 0.f, csq = 0.f;
 #pragma unroll
     for (int i = 0; i < FM; ++i) {
@@ -300,7 +300,7 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
       for (int v = 0; v < 4; ++v) {
         const int lrow = wm + i * 16 + (lane >> 4) * 4 + v;
         const long ooff = pix_out[lrow];
-        if (ooff >= 0) {
+        if (colv && ooff >= 0) {
           const float val = activate(acc[i][j][v] + bv, a.act);
           out[ooff + col] = (__bf16)val;
           csum += val;
@@ -308,6 +308,7 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
         }
       }
     }
+    // all 64 lanes reach the shuffles (no divergent early-out above)
     if (a.stats != nullptr) {
       // reduce the 4 lanes sharing this column (l, l+16, l+32, l+48), then
       // one atomic per column into a per-block bucket (64 copies cut the
@@ -316,7 +317,7 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
       csum += __shfl_xor(csum, 32, 64);
       csq += __shfl_xor(csq, 16, 64);
       csq += __shfl_xor(csq, 32, 64);
-      if ((lane >> 4) == 0) {
+      if ((lane >> 4) == 0 && colv) {
         float* bucket = a.stats + (long)(blockIdx.x & 63) * 2 * a.K;
         atomicAdd(&bucket[col], csum);
         atomicAdd(&bucket[a.K + col], csq);

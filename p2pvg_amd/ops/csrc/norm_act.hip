@@ -149,10 +149,33 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
       p2[j] += dyp * (((float)xv[j] - mn[j]) * is[j]);
     }
   }
+
+  // three-level combine so the global traffic is ONE atomic per channel per
+  // block: (1) shfl-reduce lanes sharing a channel group within the wave,
+  // (2) wave leaders combine in LDS, (3) one writer pass to global.
+  const int lane = threadIdx.x & 63;
+  extern __shared__ float sred[];  // 2*C floats
+  for (int i = threadIdx.x; i < 2 * C; i += BLOCK) sred[i] = 0.f;
+  __syncthreads();
+  if (cvec < 64) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    atomicAdd(&red[c0 + j], p1[j]);
-    atomicAdd(&red[C + c0 + j], p2[j]);
+    for (int j = 0; j < 8; ++j) {
+      for (int off = 32; off >= cvec; off >>= 1) {
+        p1[j] += __shfl_down(p1[j], off, 64);
+        p2[j] += __shfl_down(p2[j], off, 64);
+      }
+    }
+  }
+  if (lane < min(cvec, 64)) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&sred[c0 + j], p1[j]);
+      atomicAdd(&sred[C + c0 + j], p2[j]);
+    }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
+    atomicAdd(&red[i], sred[i]);
   }
 }
 
@@ -270,7 +293,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor y,
   auto stream = at::cuda::getCurrentCUDAStream();
   const long nvec = x.numel() / 8;
   hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(pick_grid(nvec)),
-                     dim3(BLOCK), 0, stream,
+                     dim3(BLOCK), 2 * C * sizeof(float), stream,
                      reinterpret_cast<const __bf16*>(x.data_ptr()),
                      reinterpret_cast<const __bf16*>(y.data_ptr()),
                      reinterpret_cast<const __bf16*>(dy.data_ptr()),
