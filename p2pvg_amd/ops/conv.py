@@ -132,7 +132,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
                     w.shape[0], w.shape[2], w.shape[3], w.shape[1]
                 ).permute(0, 3, 1, 2)
             if ctx.has_bias and ctx.needs_input_grad[2]:
-                db = g2.float().sum(0)
+                db = g2.sum(0, dtype=torch.float32)
             return dx, dw, db, None, None, None, None
 
         if ctx.needs_input_grad[0]:
@@ -149,7 +149,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
             dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
                 .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = gout.float().sum(dim=(0, 2, 3))
+            db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
         return dx, dw, db, None, None, None, None
 
 
@@ -221,7 +221,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 dwf = torch.mm(_nhwc_flat(x).t(), g2)
                 dw = dwf.view(x.shape[1], k, k, w.shape[1]).permute(0, 3, 1, 2)
             if ctx.has_bias and ctx.needs_input_grad[2]:
-                db = gout.float().sum(dim=(0, 2, 3))
+                db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
             return dx, dw, db, None, None, None, None
 
         if ctx.needs_input_grad[0]:
@@ -233,7 +233,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
             dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
                 .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = gout.float().sum(dim=(0, 2, 3))
+            db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
         return dx, dw, db, None, None, None, None
 
 
