@@ -28,9 +28,14 @@ from p2pvg_amd.models import P2PModel
 
 
 def make_synthetic_batch(cfg, device, seed: int):
-    """Synthetic BAIR-shaped clip batch: (T, B, C, H, W) in [0,1], smooth in t."""
+    """Synthetic clip batch: (T, B, C, H, W) frames in [0,1] smooth in t, or
+    (T, B, 17, 3) standardized skeletons for the h36m config."""
     g = torch.Generator(device="cpu").manual_seed(seed)
     t, b = cfg.max_seq_len, cfg.batch_size
+    if cfg.dataset == "h36m":
+        base = torch.randn(1, b, 17, 3, generator=g)
+        steps = torch.randn(t, b, 17, 3, generator=g) * 0.08
+        return (base + steps.cumsum(0)).to(device)
     c, s = cfg.channels, cfg.image_width
     base = torch.rand(1, b, c, s, s, generator=g)
     drift = torch.randn(t, b, c, 1, 1, generator=g) * 0.05
@@ -43,8 +48,9 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=32, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=256, help="per-GPU batch size")
     p.add_argument("--seq_len", type=int, default=30)
+    p.add_argument("--g_dim", type=int, default=128)
     p.add_argument("--backbone", type=str, default="vgg")
     p.add_argument("--image_width", type=int, default=64)
     p.add_argument("--dataset", type=str, default="bair")
@@ -52,7 +58,8 @@ def main():
     p.add_argument("--kernels", type=str, default="auto")
     p.add_argument("--channels_last", type=int, default=1,
                    help="NHWC activations/weights (MIOpen igemm is NHWC-native)")
-    p.add_argument("--use_graphs", type=int, default=0)
+    p.add_argument("--use_graphs", type=int, default=-1,
+                   help="-1: auto (on for single-GPU CUDA)")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", 0))
@@ -77,6 +84,7 @@ def main():
         device = torch.device("cpu")
 
     cfg = Config(
+        g_dim=args.g_dim,
         dataset=args.dataset,
         backbone=args.backbone,
         image_width=args.image_width,
@@ -89,7 +97,8 @@ def main():
         dtype=args.dtype if use_cuda else "fp32",
         device=str(device),
         ddp=world > 1,
-        use_graphs=bool(args.use_graphs) and use_cuda,
+        use_graphs=(args.use_graphs == 1 or (args.use_graphs == -1 and world == 1))
+        and use_cuda,
     )
     if args.dataset == "h36m":
         cfg.backbone = "mlp"
@@ -177,7 +186,7 @@ def main():
             "dtype": cfg.dtype,
             "data": "synthetic",
             "config": {
-                "model": f"p2pvg_{args.backbone}_{args.image_width}",
+                "model": ("p2pvg_h36m_mlp" if args.dataset == "h36m" else f"p2pvg_{args.backbone}_{args.image_width}"),
                 "dataset": args.dataset,
                 "global_batch": global_batch,
                 "seq_len": args.seq_len,
