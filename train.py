@@ -133,6 +133,12 @@ def main():
     amp_dtype = torch.bfloat16 if cfg.dtype == "bf16" else None
     qual_lengths = [10, 30]
 
+    stepper = None
+    if cfg.use_graphs and device.type == "cuda":
+        from p2pvg_amd.runtime import GraphedTrainStep
+
+        stepper = GraphedTrainStep(model, amp_dtype=amp_dtype)
+
     for epoch in range(start_epoch, cfg.nepochs):
         model.train()
         # device-side accumulators: one DtoH per log interval
@@ -142,12 +148,15 @@ def main():
 
         for i in range(cfg.epoch_size):
             x = next(train_generator)
-            model.zero_grad(set_to_none=False)
-            if amp_dtype is not None and device.type == "cuda":
-                with torch.autocast("cuda", dtype=amp_dtype):
-                    losses = model(x, 0, len(x) - 1)
+            if stepper is not None:
+                losses = stepper.step(x)
             else:
-                losses = model(x, 0, len(x) - 1)
+                model.zero_grad(set_to_none=False)
+                if amp_dtype is not None and device.type == "cuda":
+                    with torch.autocast("cuda", dtype=amp_dtype):
+                        losses = model(x, 0, len(x) - 1)
+                else:
+                    losses = model(x, 0, len(x) - 1)
             acc += torch.stack(list(losses))
             frames += (len(x) if not isinstance(x, tuple) else len(x[1])) * cfg.batch_size
 
@@ -167,6 +176,21 @@ def main():
             "[%02d] mse: %.5f | kld: %.5f | align: %.5f | cpc: %.5f | %.1f frames/s"
             % (epoch, vals[0], vals[1], vals[3], vals[2], frames / dt)
         )
+
+        # quantitative eval: end-frame SSIM (the BASELINE quality metric the
+        # reference never implemented, misc/metrics.py stub)
+        if is_main and cfg.dataset != "h36m" and (epoch + 1) % cfg.quan_iter == 0:
+            from p2pvg_amd.utils import end_frame_ssim
+
+            model.eval()
+            with torch.no_grad():
+                x = next(test_generator)
+                gen = model.p2p_generate(x, len(x), len(x) - 1, model_mode="full",
+                                         skip_frame=False)
+                s = end_frame_ssim(gen, x[len(x) - 1]).mean()
+                writer.add_scalar("Eval/end_frame_ssim", s, epoch)
+                logger.info("[%02d] end-frame SSIM (full mode): %.4f" % (epoch, s))
+            model.train()
 
         # qualitative eval (reference train.py:246-272)
         if is_main and (epoch + 1) % cfg.qual_iter == 0:
