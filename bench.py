@@ -15,6 +15,9 @@ Prints ONE JSON line from rank 0.
 """
 from __future__ import annotations
 
+import os
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import argparse
 import json
 import os

@@ -41,10 +41,10 @@ std::vector<torch::Tensor> bn_act_fwd_train(
 torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
                               torch::Tensor beta, torch::Tensor running_mean,
                               torch::Tensor running_var, double eps, long act);
-std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor y,
-                                      torch::Tensor dy, torch::Tensor mean,
-                                      torch::Tensor invstd, torch::Tensor scale,
-                                      long act);
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
+                                      torch::Tensor mean, torch::Tensor invstd,
+                                      torch::Tensor gamma, torch::Tensor beta,
+                                      torch::Tensor scale, long act);
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp);
 

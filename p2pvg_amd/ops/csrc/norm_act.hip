@@ -41,6 +41,24 @@ __device__ __forceinline__ float act_bwd_from_y(float y, int act) {
   }
 }
 
+// derivative of act given the PRE-activation value u (recomputing the cheap
+// transcendental beats re-reading the y tensor from HBM: the backward kernels
+// then touch only x and dy)
+__device__ __forceinline__ float act_bwd_from_u(float u, int act) {
+  switch (act) {
+    case 1: return u > 0.f ? 1.f : 0.2f;
+    case 2: {
+      const float t = tanhf(u);
+      return 1.f - t * t;
+    }
+    case 3: {
+      const float s = 1.f / (1.f + __expf(-u));
+      return s * (1.f - s);
+    }
+    default: return 1.f;
+  }
+}
+
 // stats (2,K) raw sums -> saved mean/invstd + scale/shift (+ running update)
 __global__ void bn_finalize_kernel(const float* __restrict__ stats,
                                    const float* __restrict__ gamma,
@@ -121,32 +139,36 @@ __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
 // FIXED across its whole grid-stride walk: partial sums live in 16 registers
 // and each thread issues exactly 16 atomics at the end.
 __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
-    const __bf16* __restrict__ x, const __bf16* __restrict__ y,
-    const __bf16* __restrict__ dy, const float* __restrict__ mean,
-    const float* __restrict__ invstd, float* __restrict__ red,  // (2,K)
+    const __bf16* __restrict__ x, const __bf16* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ red,  // (2,K)
     long nvec, int C, int act) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float p1[8], p2[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) p1[j] = p2[j] = 0.f;
-  float mn[8], is[8];
+  float mn[8], is[8], ga[8], be[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     mn[j] = mean[c0 + j];
     is[j] = invstd[c0 + j];
+    ga[j] = gamma[c0 + j];
+    be[j] = beta[c0 + j];
   }
 
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
-    bf16x8 yv = *reinterpret_cast<const bf16x8*>(y + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const float dyp = (float)gv[j] * act_bwd_from_y((float)yv[j], act);
+      const float xhat = ((float)xv[j] - mn[j]) * is[j];
+      const float dyp =
+          (float)gv[j] * act_bwd_from_u(ga[j] * xhat + be[j], act);
       p1[j] += dyp;
-      p2[j] += dyp * (((float)xv[j] - mn[j]) * is[j]);
+      p2[j] += dyp * xhat;
     }
   }
 
@@ -181,24 +203,24 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
 
 // dx = scale[c] * (dy' - s1/cnt - xhat * s2/cnt)
 __global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_kernel(
-    const __bf16* __restrict__ x, const __bf16* __restrict__ y,
-    const __bf16* __restrict__ dy, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ scale,
-    const float* __restrict__ red, __bf16* __restrict__ dx, long nvec, int C,
-    int act, float inv_count) {
+    const __bf16* __restrict__ x, const __bf16* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ scale, const float* __restrict__ red,
+    __bf16* __restrict__ dx, long nvec, int C, int act, float inv_count) {
   const int cvec = C / 8;
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
     const int c0 = (int)(i % cvec) * 8;
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
-    bf16x8 yv = *reinterpret_cast<const bf16x8*>(y + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int c = c0 + j;
-      const float dyp = (float)gv[j] * act_bwd_from_y((float)yv[j], act);
       const float xhat = ((float)xv[j] - mean[c]) * invstd[c];
+      const float dyp =
+          (float)gv[j] * act_bwd_from_u(gamma[c] * xhat + beta[c], act);
       const float v =
           scale[c] * (dyp - (red[c] + xhat * red[C + c]) * inv_count);
       o[j] = (__bf16)v;
@@ -280,11 +302,12 @@ torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
   return y;
 }
 
-// Backward: returns (dx, dgamma, dbeta).
-std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor y,
-                                      torch::Tensor dy, torch::Tensor mean,
-                                      torch::Tensor invstd, torch::Tensor scale,
-                                      long act) {
+// Backward: returns (dx, dgamma, dbeta). Reads only x and dy (activation
+// derivative recomputed from the pre-activation value).
+std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
+                                      torch::Tensor mean, torch::Tensor invstd,
+                                      torch::Tensor gamma, torch::Tensor beta,
+                                      torch::Tensor scale, long act) {
   const int C = x.size(1);
   const long count = x.numel() / C;
   auto f32 = x.options().dtype(torch::kFloat32);
@@ -295,16 +318,16 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor y,
   hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(pick_grid(nvec)),
                      dim3(BLOCK), 2 * C * sizeof(float), stream,
                      reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     reinterpret_cast<const __bf16*>(y.data_ptr()),
                      reinterpret_cast<const __bf16*>(dy.data_ptr()),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      red.data_ptr<float>(), nvec, C, (int)act);
   hipLaunchKernelGGL(bn_act_bwd_apply_kernel, dim3(pick_grid(nvec)),
                      dim3(BLOCK), 0, stream,
                      reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     reinterpret_cast<const __bf16*>(y.data_ptr()),
                      reinterpret_cast<const __bf16*>(dy.data_ptr()),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      scale.data_ptr<float>(), red.data_ptr<float>(),
                      reinterpret_cast<__bf16*>(dx.data_ptr()), nvec, C,
                      (int)act, (float)(1.0 / count));

@@ -44,22 +44,26 @@ class FusedBNActFn(torch.autograd.Function):
     def forward(ctx, x, stats, gamma, beta, running_mean, running_var,
                 momentum: float, eps: float, act: int):
         ext = _ext()
+        gamma32 = gamma.float()
+        beta32 = beta.float()
         y, mean, invstd, scale = ext.bn_act_fwd_train(
-            x, stats, gamma.float(), beta.float(), running_mean, running_var,
+            x, stats, gamma32, beta32, running_mean, running_var,
             momentum, eps, act,
         )
-        ctx.save_for_backward(x, y, mean, invstd, scale)
+        ctx.save_for_backward(x, mean, invstd, gamma32, beta32, scale)
         ctx.act = act
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, mean, invstd, scale = ctx.saved_tensors
+        x, mean, invstd, gamma32, beta32, scale = ctx.saved_tensors
         ext = _ext()
         dy = dy.contiguous(memory_format=CL)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
-        dx, dgamma, dbeta = ext.bn_act_bwd(x, y, dy, mean, invstd, scale, ctx.act)
+        dx, dgamma, dbeta = ext.bn_act_bwd(
+            x, dy, mean, invstd, gamma32, beta32, scale, ctx.act
+        )
         return dx, None, dgamma, dbeta, None, None, None, None, None
 
 
@@ -73,13 +77,17 @@ def fused_conv_bn_act(x, conv, bn, act: int):
         xl = _to_cl_bf16(x)
         wl = _to_cl_bf16(conv.weight)
         want_stats = training
+        # BatchNorm is shift-invariant, so the conv bias has EXACTLY zero
+        # effect on the block output; skip it (and its gradient reduction).
+        # The reference trains this bias on float-rounding noise only — it is
+        # initialized to 0 (init_weights) and stays ~0.
         if isinstance(conv, nn.ConvTranspose2d):
             out, stats = ConvT2dNHWCFn.apply(
-                xl, wl, conv.bias, conv.stride[0], conv.padding[0], 0, want_stats
+                xl, wl, None, conv.stride[0], conv.padding[0], 0, want_stats
             )
         else:
             out, stats = Conv2dNHWCFn.apply(
-                xl, wl, conv.bias, conv.stride[0], conv.padding[0], 0, want_stats
+                xl, wl, None, conv.stride[0], conv.padding[0], 0, want_stats
             )
         if not want_stats:
             stats = None
