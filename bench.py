@@ -17,6 +17,7 @@ from __future__ import annotations
 
 import os
 os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+os.environ.setdefault("PYTORCH_HIP_ALLOC_CONF", "expandable_segments:True")
 
 import argparse
 import json
