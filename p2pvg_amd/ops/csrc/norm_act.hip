@@ -115,19 +115,27 @@ __global__ void bn_eval_prep_kernel(const float* __restrict__ running_mean,
 }
 
 // y = act(x*scale[c] + shift[c]); vectors of 8 bf16 along C
+// C/8 divides BLOCK and the grid stride, so each thread's channel group is
+// fixed: per-channel scalars are preloaded into registers once.
 __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
     const __bf16* __restrict__ x, const float* __restrict__ scale,
     const float* __restrict__ shift, __bf16* __restrict__ y, long nvec, int C,
     int act) {
   const int cvec = C / 8;
+  const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = scale[c0 + j];
+    sh[j] = shift[c0 + j];
+  }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    const int c0 = (int)(i % cvec) * 8;
     bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const float vv = (float)v[j] * scale[c0 + j] + shift[c0 + j];
+      const float vv = (float)v[j] * sc[j] + sh[j];
       o[j] = (__bf16)act_fwd(vv, act);
     }
     *reinterpret_cast<bf16x8*>(y + i * 8) = o;
@@ -209,20 +217,29 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_kernel(
     const float* __restrict__ scale, const float* __restrict__ red,
     __bf16* __restrict__ dx, long nvec, int C, int act, float inv_count) {
   const int cvec = C / 8;
+  const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
+  float mn[8], is[8], ga[8], be[8], sc[8], r1[8], r2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mn[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+    ga[j] = gamma[c0 + j];
+    be[j] = beta[c0 + j];
+    sc[j] = scale[c0 + j];
+    r1[j] = red[c0 + j] * inv_count;
+    r2[j] = red[C + c0 + j] * inv_count;
+  }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    const int c0 = (int)(i % cvec) * 8;
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int c = c0 + j;
-      const float xhat = ((float)xv[j] - mean[c]) * invstd[c];
+      const float xhat = ((float)xv[j] - mn[j]) * is[j];
       const float dyp =
-          (float)gv[j] * act_bwd_from_u(gamma[c] * xhat + beta[c], act);
-      const float v =
-          scale[c] * (dyp - (red[c] + xhat * red[C + c]) * inv_count);
+          (float)gv[j] * act_bwd_from_u(ga[j] * xhat + be[j], act);
+      const float v = sc[j] * (dyp - r1[j] - xhat * r2[j]);
       o[j] = (__bf16)v;
     }
     *reinterpret_cast<bf16x8*>(dx + i * 8) = o;
