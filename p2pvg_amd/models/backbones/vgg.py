@@ -13,6 +13,7 @@ import torch.nn as nn
 from ...ops.conv import Conv2d, ConvTranspose2d
 from ...ops.norm import BatchNorm2d
 from ...ops.fused_norm import FusedSequential
+from ...ops.pool import MaxPool2d, UpsamplingNearest2d
 
 from .blocks import vgg_layer
 
@@ -32,7 +33,7 @@ class Encoder64(nn.Module):
         self.c5 = FusedSequential(
             Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
         )
-        self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
+        self.mp = MaxPool2d(kernel_size=2, stride=2, padding=0)
 
     def forward(self, x):
         h1 = self.c1(x)
@@ -64,7 +65,7 @@ class Decoder64(nn.Module):
             ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
         )
-        self.up = nn.UpsamplingNearest2d(scale_factor=2)
+        self.up = UpsamplingNearest2d(scale_factor=2)
 
     def forward(self, inp):
         vec, skip = inp
@@ -93,7 +94,7 @@ class Encoder128(nn.Module):
         self.c6 = FusedSequential(
             Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
         )
-        self.mp = nn.MaxPool2d(kernel_size=2, stride=2, padding=0)
+        self.mp = MaxPool2d(kernel_size=2, stride=2, padding=0)
 
     def forward(self, x):
         h1 = self.c1(x)
@@ -129,7 +130,7 @@ class Decoder128(nn.Module):
             ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
         )
-        self.up = nn.UpsamplingNearest2d(scale_factor=2)
+        self.up = UpsamplingNearest2d(scale_factor=2)
 
     def forward(self, inp):
         vec, skip = inp

@@ -47,6 +47,11 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                                       torch::Tensor scale, long act);
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp);
+std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor in);
+torch::Tensor maxpool2x2_bwd(torch::Tensor gout, torch::Tensor idx, long H,
+                             long W);
+torch::Tensor upsample2x_fwd(torch::Tensor in);
+torch::Tensor upsample2x_bwd(torch::Tensor gout);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
@@ -66,6 +71,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act bwd (gfx950)");
   m.def("conv2d_nhwc_wgrad", &conv2d_nhwc_wgrad,
         "NHWC wgrad, split-K over pixels, fp32 workspace (gfx950 MFMA)");
+  m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC 2x2/s2 maxpool fwd (gfx950)");
+  m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC 2x2/s2 maxpool bwd (gfx950)");
+  m.def("upsample2x_fwd", &upsample2x_fwd, "NHWC nearest x2 fwd (gfx950)");
+  m.def("upsample2x_bwd", &upsample2x_bwd, "NHWC nearest x2 bwd (gfx950)");
   m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (gfx950)");
   m.def("lstm_cell_bwd_pointwise", &lstm_cell_bwd_pointwise,
         "LSTM cell backward pointwise (gfx950)");
