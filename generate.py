@@ -58,15 +58,42 @@ def read_image_pair(start_img: str, end_img: str) -> torch.Tensor:
     return torch.stack(out).unsqueeze(1)
 
 
+def multi_cp_generate(model, control_frames, seg_len: int):
+    """Chain point-to-point segments through a list of control frames,
+    carrying hidden state across segments (the reference README's
+    'multiple control points generation' capability, realized through
+    p2p_generate's init_hidden flag)."""
+    out = []
+    init = True
+    for a, b in zip(control_frames[:-1], control_frames[1:]):
+        seq = torch.stack([a, b])  # (2, B, C, H, W)
+        gen = model.p2p_generate(seq, seg_len, seg_len - 1, model_mode="full",
+                                 init_hidden=init)
+        init = False
+        out.extend(gen if not out else gen[1:])
+    return out
+
+
+def loop_generate(model, start_frame, end_frame, seg_len: int):
+    """A -> B -> A loop (the reference README's 'loop generation')."""
+    return multi_cp_generate(model, [start_frame, end_frame, start_frame], seg_len)
+
+
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--ckpt", type=str, required=True, help="model .pth file")
-    parser.add_argument("--video", type=str, default="", help=".mp4/.gif video or frame dir")
+    parser.add_argument("--video", type=str, default="", help=".gif video or frame dir")
     parser.add_argument("--start_img", type=str, default="")
     parser.add_argument("--end_img", type=str, default="")
     parser.add_argument("--output_root", type=str, default="gen_outputs")
     parser.add_argument("--seed", type=int, default=1)
     parser.add_argument("--device", type=str, default="auto")
+    parser.add_argument("--multi_cp", action="store_true",
+                        help="chain segments through every input frame as a control point")
+    parser.add_argument("--loop", action="store_true",
+                        help="loop generation: first -> last -> first")
+    parser.add_argument("--segment_len", type=int, default=10,
+                        help="frames per segment for --multi_cp / --loop")
     args = parser.parse_args()
 
     states = load_checkpoint(args.ckpt)
@@ -102,6 +129,26 @@ def main():
     seq_len = len(seq)
 
     os.makedirs(args.output_root, exist_ok=True)
+
+    if args.multi_cp or args.loop:
+        with torch.no_grad():
+            if args.loop:
+                gen = loop_generate(model, seq[0], seq[seq_len - 1], args.segment_len)
+                stem = "loopgen"
+            else:
+                gen = multi_cp_generate(model, list(seq), args.segment_len)
+                stem = "mulcpgen"
+            gen = torch.stack(gen).cpu().float()
+            save_image(
+                make_grid(gen[:, 0], nrow=len(gen), padding=0),
+                os.path.join(args.output_root, f"{stem}.png"),
+            )
+            save_gif(
+                os.path.join(args.output_root, f"{stem}.gif"),
+                [to_uint8_hwc(gen[t, 0]) for t in range(len(gen))],
+            )
+            print(f"[*] wrote {stem} outputs ({len(gen)} frames) to {args.output_root}")
+        return
 
     with torch.no_grad():
         for length_to_gen in gen_lengths:

@@ -36,21 +36,22 @@ __device__ __forceinline__ int swzp(int ch, int pix) {
   return pix ^ (((ch >> 3) & 7) << 3);
 }
 
-// BT: channel tile on both sides (64 or 128); wave tile BT/2, BT/32 frags.
-template <int BT>
+// BTB/BTA: channel tiles per side (64 or 128, independent); wave tile
+// (BTB/2 x BTA/2), (BTB/32 x BTA/32) fragments per wave.
+template <int BTB, int BTA>
 __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     const __bf16* __restrict__ Y,  // (N, HO, WO, B)
     const __bf16* __restrict__ X,  // (N, H, W, A)
     float* __restrict__ ws,        // (B, R, S, A) fp32, pre-zeroed
     int Nb, int HO, int WO, int B, int H, int W, int A, int R, int S,
     int STRIDE, int PAD, int p_per_slab) {
-  __shared__ __align__(16) char lds[2 * BT * ROWB];
-  char* yt = lds;                       // [BT rows (b)][PIT pix]
-  char* xt = lds + BT * ROWB;           // [BT rows (a)][PIT pix]
+  __shared__ __align__(16) char lds[(BTB + BTA) * ROWB];
+  char* yt = lds;                       // [BTB rows (b)][PIT pix]
+  char* xt = lds + BTB * ROWB;          // [BTA rows (a)][PIT pix]
 
-  const int at_blocks = (A + BT - 1) / BT;
-  const int b0 = (blockIdx.x / at_blocks) * BT;
-  const int a0 = (blockIdx.x % at_blocks) * BT;
+  const int at_blocks = (A + BTA - 1) / BTA;
+  const int b0 = (blockIdx.x / at_blocks) * BTB;
+  const int a0 = (blockIdx.x % at_blocks) * BTA;
   const int r = blockIdx.y / S;
   const int s = blockIdx.y % S;
   const int p_begin = blockIdx.z * p_per_slab;
@@ -60,45 +61,56 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
-  constexpr int FRG = BT / 32;
-  const int wm = (wid >> 1) * (BT / 2);   // wave row (b) base
-  const int wn = (wid & 1) * (BT / 2);    // wave col (a) base
+  constexpr int FRB = BTB / 32;
+  constexpr int FRA = BTA / 32;
+  const int wm = (wid >> 1) * (BTB / 2);   // wave row (b) base
+  const int wn = (wid & 1) * (BTA / 2);    // wave col (a) base
 
-  f32x4 acc[FRG][FRG];
+  f32x4 acc[FRB][FRA];
 #pragma unroll
-  for (int i = 0; i < FRG; ++i)
+  for (int i = 0; i < FRB; ++i)
 #pragma unroll
-    for (int j = 0; j < FRG; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FRA; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  constexpr int SL = (PCH * BT / 8) / THREADS;  // staging slots per thread
-  bf16x8 yreg[SL], xreg[SL];
+  constexpr int SLY = (PCH * BTB / 8) / THREADS;  // Y staging slots/thread
+  constexpr int SLX = (PCH * BTA / 8) / THREADS;  // X staging slots/thread
+  bf16x8 yreg[SLY], xreg[SLX];
 
   // T14 pipeline: issue chunk t+1's global loads, MFMA chunk t from LDS,
   // write t+1 after the barrier. Pixel meta is computed inline per staging
   // slot (few int divides, hidden under the loads).
   auto load_chunk = [&](int p0) {
 #pragma unroll
-    for (int it = 0; it < SL; ++it) {
+    for (int it = 0; it < SLY; ++it) {
       const int slot = it * THREADS + tid;
-      const int pix_l = slot / (BT / 8);
-      const int ch0 = (slot % (BT / 8)) * 8;
+      const int pix_l = slot / (BTB / 8);
+      const int ch0 = (slot % (BTB / 8)) * 8;
       const int pix = p0 + pix_l;
-      bf16x8 vy = {}, vx = {};
+      bf16x8 vy = {};
+      if (pix < p_end && b0 + ch0 < B) {
+        const __bf16* src = Y + (long)pix * B + b0 + ch0;
+        if (b0 + ch0 + 8 <= B) {
+          vy = *reinterpret_cast<const bf16x8*>(src);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            if (b0 + ch0 + j < B) vy[j] = src[j];
+        }
+      }
+      yreg[it] = vy;
+    }
+#pragma unroll
+    for (int it = 0; it < SLX; ++it) {
+      const int slot = it * THREADS + tid;
+      const int pix_l = slot / (BTA / 8);
+      const int ch0 = (slot % (BTA / 8)) * 8;
+      const int pix = p0 + pix_l;
+      bf16x8 vx = {};
       if (pix < p_end) {
         const int n = pix / (HO * WO);
         const int rem = pix - n * (HO * WO);
         const int ho = rem / WO;
         const int wo = rem - ho * WO;
-        if (b0 + ch0 < B) {
-          const __bf16* src = Y + (long)pix * B + b0 + ch0;
-          if (b0 + ch0 + 8 <= B) {
-            vy = *reinterpret_cast<const bf16x8*>(src);
-          } else {
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-              if (b0 + ch0 + j < B) vy[j] = src[j];
-          }
-        }
         const int hi = ho * STRIDE - PAD + r;
         const int wi = wo * STRIDE - PAD + s;
         if (hi >= 0 && hi < H && wi >= 0 && wi < W && a0 + ch0 < A) {
@@ -113,7 +125,6 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
           }
         }
       }
-      yreg[it] = vy;
       xreg[it] = vx;
     }
   };
@@ -122,15 +133,22 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     __bf16* yd = reinterpret_cast<__bf16*>(yt);
     __bf16* xd = reinterpret_cast<__bf16*>(xt);
 #pragma unroll
-    for (int it = 0; it < SL; ++it) {
+    for (int it = 0; it < SLY; ++it) {
       const int slot = it * THREADS + tid;
-      const int pix_l = slot / (BT / 8);
-      const int ch0 = (slot % (BT / 8)) * 8;
+      const int pix_l = slot / (BTB / 8);
+      const int ch0 = (slot % (BTB / 8)) * 8;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
+      for (int j = 0; j < 8; ++j)
         yd[(ch0 + j) * PIT + swzp(ch0 + j, pix_l)] = yreg[it][j];
+    }
+#pragma unroll
+    for (int it = 0; it < SLX; ++it) {
+      const int slot = it * THREADS + tid;
+      const int pix_l = slot / (BTA / 8);
+      const int ch0 = (slot % (BTA / 8)) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
         xd[(ch0 + j) * PIT + swzp(ch0 + j, pix_l)] = xreg[it][j];
-      }
     }
   };
 
@@ -147,20 +165,23 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int pe = kk * 32 + (lane >> 4) * 8;   // pixel base (8-aligned)
-      bf16x8 a_frag[FRG], b_frag[FRG];
+      bf16x8 a_frag[FRB], b_frag[FRA];
 #pragma unroll
-      for (int f = 0; f < FRG; ++f) {
+      for (int f = 0; f < FRB; ++f) {
         const int brow = wm + f * 16 + (lane & 15);
         a_frag[f] = *reinterpret_cast<const bf16x8*>(
             yt + brow * ROWB + swzp(brow, pe) * 2);
+      }
+#pragma unroll
+      for (int f = 0; f < FRA; ++f) {
         const int arow = wn + f * 16 + (lane & 15);
         b_frag[f] = *reinterpret_cast<const bf16x8*>(
             xt + arow * ROWB + swzp(arow, pe) * 2);
       }
 #pragma unroll
-      for (int i = 0; i < FRG; ++i)
+      for (int i = 0; i < FRB; ++i)
 #pragma unroll
-        for (int j = 0; j < FRG; ++j)
+        for (int j = 0; j < FRA; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
     }
@@ -174,11 +195,11 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 
   // epilogue: atomic accumulate into the fp32 workspace
 #pragma unroll
-  for (int j = 0; j < FRG; ++j) {
+  for (int j = 0; j < FRA; ++j) {
     const int a = a0 + wn + j * 16 + (lane & 15);
     if (a >= A) continue;
 #pragma unroll
-    for (int i = 0; i < FRG; ++i) {
+    for (int i = 0; i < FRB; ++i) {
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
         const int b = b0 + wm + i * 16 + (lane >> 4) * 4 + v;
@@ -209,8 +230,9 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
   auto ws = torch::zeros({B, (long)R, (long)S, A},
                          Y.options().dtype(torch::kFloat32));
 
-  const int BTsel = (B >= 128 && A >= 128) ? 128 : 64;
-  const int bt = ceil_div(B, BTsel), at = ceil_div(A, BTsel);
+  const int BTB = B >= 128 ? 128 : 64;
+  const int BTA = A >= 128 ? 128 : 64;
+  const int bt = ceil_div(B, BTB), at = ceil_div(A, BTA);
   const int p_total = Nb * HO * WO;
   int sp = (int)splitp;
   if (sp <= 0) {
@@ -223,18 +245,16 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
 
   dim3 grid(bt * at, (int)(R * S), sp);
   auto stream = at::cuda::getCurrentCUDAStream();
-  if (BTsel == 128) {
-    hipLaunchKernelGGL(conv2d_wgrad_kernel<128>, grid, dim3(THREADS), 0, stream,
-                       reinterpret_cast<const __bf16*>(Y.data_ptr()),
-                       reinterpret_cast<const __bf16*>(X.data_ptr()),
-                       ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
-                       (int)S, (int)stride, (int)pad, p_per_slab);
-  } else {
-    hipLaunchKernelGGL(conv2d_wgrad_kernel<64>, grid, dim3(THREADS), 0, stream,
-                       reinterpret_cast<const __bf16*>(Y.data_ptr()),
-                       reinterpret_cast<const __bf16*>(X.data_ptr()),
-                       ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
-                       (int)S, (int)stride, (int)pad, p_per_slab);
-  }
+#define WGRAD_LAUNCH(BB, AA)                                                   \
+  hipLaunchKernelGGL((conv2d_wgrad_kernel<BB, AA>), grid, dim3(THREADS), 0,    \
+                     stream, reinterpret_cast<const __bf16*>(Y.data_ptr()),    \
+                     reinterpret_cast<const __bf16*>(X.data_ptr()),            \
+                     ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,     \
+                     (int)S, (int)stride, (int)pad, p_per_slab)
+  if (BTB == 128 && BTA == 128) WGRAD_LAUNCH(128, 128);
+  else if (BTB == 128) WGRAD_LAUNCH(128, 64);
+  else if (BTA == 128) WGRAD_LAUNCH(64, 128);
+  else WGRAD_LAUNCH(64, 64);
+#undef WGRAD_LAUNCH
   return ws;
 }

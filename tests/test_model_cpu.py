@@ -111,3 +111,22 @@ def test_all_backbones_shapes(backbone, width):
     assert latent.shape == (2, 32)
     out = dec([latent, skips])
     assert out.shape == x.shape
+
+
+def test_multi_cp_and_loop_generation(tiny_cfg):
+    """Chained segment generation through multiple control points and the
+    A->B->A loop (reference README showcase capabilities)."""
+    import generate as gen_cli
+
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(tiny_cfg)
+    model.eval()
+    x = make_batch(tiny_cfg)
+    frames = [x[0], x[3], x[7]]
+    out = gen_cli.multi_cp_generate(model, frames, seg_len=5)
+    # two segments of 5 sharing the joint frame
+    assert len(out) == 9
+    loop = gen_cli.loop_generate(model, x[0], x[7], seg_len=4)
+    assert len(loop) == 7
+    assert torch.equal(loop[0], x[0])
