@@ -4,7 +4,7 @@ import sys
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, "/root/repo")
 from p2pvg_amd.ops import _hip_ext_loader
 
 ext = _hip_ext_loader.load()
