@@ -24,6 +24,7 @@ std::vector<torch::Tensor> gaussian_kl_bwd(torch::Tensor mu1, torch::Tensor lv1,
                                            torch::Tensor dout, double denom);
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+torch::Tensor tr16_probe(long mode);
 torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
                               c10::optional<torch::Tensor> bias, long stride,
                               long pad, long act,
@@ -55,6 +56,7 @@ torch::Tensor upsample2x_bwd(torch::Tensor gout);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd,
         "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA)",
         pybind11::arg("in"), pybind11::arg("w"), pybind11::arg("bias"),
