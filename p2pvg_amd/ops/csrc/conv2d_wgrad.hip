@@ -11,10 +11,11 @@
 // atomicAdd into a workspace (deterministic mode: splitP=1).
 //
 // Both operands are pixel-major in memory (NHWC), so global loads are
-// contiguous 16B channel runs; tiles are TRANSPOSED while staging into LDS
-// ([channel][pixel], odd row pitch so the b128 fragment reads are spread
-// across banks), which makes every MFMA fragment read a contiguous
-// ds_read_b128 along the pixel (reduction) axis.
+// contiguous 16B channel runs and staging writes are single ds_write_b128s
+// into [pix/4][ch/16][4][16] subtiles; the transpose to reduction-axis
+// fragments happens IN HARDWARE via ds_read_b64_tr_b16 (gfx950's LDS
+// transpose read: a 16-lane group reads one 128-byte [4pix][16ch] block and
+// lane c receives column c — semantics pinned by tests/test_tr16_gpu.py).
 
 #include "common.h"
 
@@ -25,15 +26,33 @@ namespace {
 
 constexpr int PCH = 64;    // pixels per staging chunk
 constexpr int THREADS = 256;
-constexpr int PIT = PCH + 8;          // LDS row pitch in elements (16B-aligned)
-constexpr int ROWB = PIT * 2;         // bytes per LDS row
 
-// pixel-position XOR swizzle: without it, the transposed staging writes of
-// 16 lanes land on one bank (channel stride 8*PIT*2 B = 0 mod 32 banks,
-// 16-way conflict). XORing the pixel slot by bits of the channel spreads
-// them; reads use the same XOR on their 8-aligned pixel base (guide T2).
-__device__ __forceinline__ int swzp(int ch, int pix) {
-  return pix ^ (((ch >> 3) & 7) << 3);
+typedef unsigned short u16x4 __attribute__((ext_vector_type(4)));
+
+// [pix/4][BT/16][4][16] subtiled LDS image: byte offset of element (pix, ch)
+template <int BT>
+__device__ __forceinline__ int sub_off(int pix, int ch) {
+  return ((pix >> 2) * (BT / 16) + (ch >> 4)) * 128 + (pix & 3) * 32 +
+         (ch & 15) * 2;
+}
+
+// two hardware transpose reads assembling the 8-pixel-run fragment for a
+// fixed channel: lane l covers ch (..&15 == l&15), pixels pe..pe+7
+__device__ __forceinline__ bf16x8 tr16_frag(const char* tile, int ch, int pe,
+                                            int lane, int bt_over_16) {
+  const unsigned a0 =
+      (unsigned)(unsigned long long)tile +
+      (unsigned)(((pe >> 2) * bt_over_16 + (ch >> 4)) * 128 + (lane & 15) * 8);
+  const unsigned a1 = a0 + (unsigned)(bt_over_16 * 128);
+  u16x4 lo, hi;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(lo) : "v"(a0));
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(hi) : "v"(a1));
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  u16x4 l2 = lo, h2 = hi;
+  typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
+  u16x8 r = __builtin_shufflevector(l2, h2, 0, 1, 2, 3, 4, 5, 6, 7);
+  return __builtin_bit_cast(bf16x8, r);
 }
 
 // BTB/BTA: channel tiles per side (64 or 128, independent); wave tile
@@ -45,9 +64,9 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     float* __restrict__ ws,        // (B, R, S, A) fp32, pre-zeroed
     int Nb, int HO, int WO, int B, int H, int W, int A, int R, int S,
     int STRIDE, int PAD, int p_per_slab) {
-  __shared__ __align__(16) char lds[(BTB + BTA) * ROWB];
-  char* yt = lds;                       // [BTB rows (b)][PIT pix]
-  char* xt = lds + BTB * ROWB;          // [BTA rows (a)][PIT pix]
+  __shared__ __align__(16) char lds[PCH * (BTB + BTA) * 2];
+  char* yt = lds;                        // [PCH/4][BTB/16][4][16] subtiles
+  char* xt = lds + PCH * BTB * 2;        // [PCH/4][BTA/16][4][16] subtiles
 
   const int at_blocks = (A + BTA - 1) / BTA;
   const int b0 = (blockIdx.x / at_blocks) * BTB;
@@ -130,25 +149,19 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
   };
 
   auto write_chunk = [&]() {
-    __bf16* yd = reinterpret_cast<__bf16*>(yt);
-    __bf16* xd = reinterpret_cast<__bf16*>(xt);
 #pragma unroll
     for (int it = 0; it < SLY; ++it) {
       const int slot = it * THREADS + tid;
       const int pix_l = slot / (BTB / 8);
       const int ch0 = (slot % (BTB / 8)) * 8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        yd[(ch0 + j) * PIT + swzp(ch0 + j, pix_l)] = yreg[it][j];
+      *reinterpret_cast<bf16x8*>(yt + sub_off<BTB>(pix_l, ch0)) = yreg[it];
     }
 #pragma unroll
     for (int it = 0; it < SLX; ++it) {
       const int slot = it * THREADS + tid;
       const int pix_l = slot / (BTA / 8);
       const int ch0 = (slot % (BTA / 8)) * 8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        xd[(ch0 + j) * PIT + swzp(ch0 + j, pix_l)] = xreg[it][j];
+      *reinterpret_cast<bf16x8*>(xt + sub_off<BTA>(pix_l, ch0)) = xreg[it];
     }
   };
 
@@ -169,14 +182,12 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 #pragma unroll
       for (int f = 0; f < FRB; ++f) {
         const int brow = wm + f * 16 + (lane & 15);
-        a_frag[f] = *reinterpret_cast<const bf16x8*>(
-            yt + brow * ROWB + swzp(brow, pe) * 2);
+        a_frag[f] = tr16_frag(yt, brow, pe, lane, BTB / 16);
       }
 #pragma unroll
       for (int f = 0; f < FRA; ++f) {
         const int arow = wn + f * 16 + (lane & 15);
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(
-            xt + arow * ROWB + swzp(arow, pe) * 2);
+        b_frag[f] = tr16_frag(xt, arow, pe, lane, BTA / 16);
       }
 #pragma unroll
       for (int i = 0; i < FRB; ++i)

@@ -14,17 +14,26 @@ __global__ void tr16_probe_kernel(unsigned short* __restrict__ out, int mode) {
   const int l = threadIdx.x & 63;
   for (int i = l; i < 1024; i += 64) lds[i] = (unsigned short)i;
   __syncthreads();
-  unsigned addr;
-  switch (mode) {
-    case 0: addr = l * 8; break;                    // each lane its own 8B
-    case 1: addr = (l & 15) * 8 + (l >> 4) * 128; break;
-    case 2: addr = (l >> 4) * 128; break;           // uniform within group
-    default: addr = 0; break;                       // fully uniform
+  const unsigned base = (unsigned)(unsigned long long)(&lds[0]);
+  unsigned off;
+  switch (mode & 3) {
+    case 0: off = l * 8; break;                     // each lane its own 8B
+    case 1: off = (l & 15) * 8 + (l >> 4) * 128; break;
+    case 2: off = (l >> 4) * 128; break;            // uniform within group
+    default: off = 0; break;                        // fully uniform
   }
+  const unsigned addr = base + off;
   u16x4 v;
-  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
-               : "=v"(v)
-               : "v"(addr));
+  if (mode >= 4) {
+    // control: plain ds_read_b64 at the same address
+    asm volatile("ds_read_b64 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                 : "=v"(v)
+                 : "v"(addr));
+  } else {
+    asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                 : "=v"(v)
+                 : "v"(addr));
+  }
   __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
   for (int j = 0; j < 4; ++j) out[l * 4 + j] = v[j];
