@@ -36,22 +36,22 @@ __device__ __forceinline__ int sub_off(int pix, int ch) {
          (ch & 15) * 2;
 }
 
-// two hardware transpose reads assembling the 8-pixel-run fragment for a
-// fixed channel: lane l covers ch (..&15 == l&15), pixels pe..pe+7
-__device__ __forceinline__ bf16x8 tr16_frag(const char* tile, int ch, int pe,
-                                            int lane, int bt_over_16) {
+// issue one pair of hardware transpose reads (no wait) for the 8-pixel-run
+// fragment of a fixed channel: lane l covers ch (..&15 == l&15), pix pe..pe+7
+__device__ __forceinline__ void tr16_issue(const char* tile, int ch, int pe,
+                                           int lane, int bt_over_16,
+                                           u16x4& lo, u16x4& hi) {
   const unsigned a0 =
       (unsigned)(unsigned long long)tile +
       (unsigned)(((pe >> 2) * bt_over_16 + (ch >> 4)) * 128 + (lane & 15) * 8);
   const unsigned a1 = a0 + (unsigned)(bt_over_16 * 128);
-  u16x4 lo, hi;
   asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(lo) : "v"(a0));
   asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(hi) : "v"(a1));
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_sched_barrier(0);
-  u16x4 l2 = lo, h2 = hi;
+}
+
+__device__ __forceinline__ bf16x8 tr16_combine(u16x4 lo, u16x4 hi) {
   typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
-  u16x8 r = __builtin_shufflevector(l2, h2, 0, 1, 2, 3, 4, 5, 6, 7);
+  u16x8 r = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
   return __builtin_bit_cast(bf16x8, r);
 }
 
@@ -178,17 +178,22 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int pe = kk * 32 + (lane >> 4) * 8;   // pixel base (8-aligned)
+      u16x4 alo[FRB], ahi[FRB], blo[FRA], bhi[FRA];
+#pragma unroll
+      for (int f = 0; f < FRB; ++f)
+        tr16_issue(yt, wm + f * 16 + (lane & 15), pe, lane, BTB / 16, alo[f],
+                   ahi[f]);
+#pragma unroll
+      for (int f = 0; f < FRA; ++f)
+        tr16_issue(xt, wn + f * 16 + (lane & 15), pe, lane, BTA / 16, blo[f],
+                   bhi[f]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
       bf16x8 a_frag[FRB], b_frag[FRA];
 #pragma unroll
-      for (int f = 0; f < FRB; ++f) {
-        const int brow = wm + f * 16 + (lane & 15);
-        a_frag[f] = tr16_frag(yt, brow, pe, lane, BTB / 16);
-      }
+      for (int f = 0; f < FRB; ++f) a_frag[f] = tr16_combine(alo[f], ahi[f]);
 #pragma unroll
-      for (int f = 0; f < FRA; ++f) {
-        const int arow = wn + f * 16 + (lane & 15);
-        b_frag[f] = tr16_frag(xt, arow, pe, lane, BTA / 16);
-      }
+      for (int f = 0; f < FRA; ++f) b_frag[f] = tr16_combine(blo[f], bhi[f]);
 #pragma unroll
       for (int i = 0; i < FRB; ++i)
 #pragma unroll
