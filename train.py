@@ -135,9 +135,13 @@ def main():
 
     stepper = None
     if cfg.use_graphs and device.type == "cuda":
-        from p2pvg_amd.runtime import GraphedTrainStep
+        if world > 1:
+            logger.info("[!] hipGraph capture with RCCL collectives is untested; "
+                        "running DDP eager (graphs disabled)")
+        else:
+            from p2pvg_amd.runtime import GraphedTrainStep
 
-        stepper = GraphedTrainStep(model, amp_dtype=amp_dtype)
+            stepper = GraphedTrainStep(model, amp_dtype=amp_dtype)
 
     for epoch in range(start_epoch, cfg.nepochs):
         model.train()
