@@ -32,12 +32,12 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
 
-constexpr int BK = 64;              // chunk width (bf16 elements)
 constexpr int THREADS = 256;
-constexpr int ROW_BYTES = BK * 2;   // 128 B per LDS tile row
 
+// BK = K-chunk width in bf16 elements (LDS tile row length)
+template <int BK>
 __device__ __forceinline__ int swz(int row, int cb) {
-  return row * ROW_BYTES + (cb ^ ((row & 7) << 4));
+  return row * (BK * 2) + (cb ^ ((row & 7) << 4));
 }
 
 __device__ __forceinline__ float activate(float v, int act) {
@@ -67,14 +67,14 @@ struct ConvArgs {
 
 // KSIZE: compact kernel size (taps per axis). FRAC: parity mode. RSCLIN:
 // rsc-linear chunking for small C.
-template <int BM, int BN, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
+template <int BM, int BN, int BK, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
 __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ w,
     const float* __restrict__ bias, __bf16* __restrict__ out, ConvArgs a) {
-  __shared__ __align__(16) char lds[(BM + BN) * ROW_BYTES + BM * 16];
+  __shared__ __align__(16) char lds[(BM + BN) * (BK * 2) + BM * 16];
   char* a_lds = lds;
-  char* b_lds = lds + BM * ROW_BYTES;
-  long* pix_out = reinterpret_cast<long*>(lds + (BM + BN) * ROW_BYTES);
+  char* b_lds = lds + BM * (BK * 2);
+  long* pix_out = reinterpret_cast<long*>(lds + (BM + BN) * (BK * 2));
   int* pix_off = reinterpret_cast<int*>(pix_out + BM);
   short* pix_hi = reinterpret_cast<short*>(pix_off + BM);
   short* pix_wi = pix_hi + BM;
@@ -137,8 +137,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
   const int n_inner = RSCLIN ? (RSC + BK - 1) / BK : (a.C + BK - 1) / BK;
   const int nchunks = n_outer * n_inner;
 
-  constexpr int ASL = (BM * 8) / THREADS;   // A staging slots per thread
-  constexpr int BSL = (BN * 8) / THREADS;   // B staging slots per thread
+  constexpr int ASL = (BM * (BK / 8)) / THREADS;   // A staging slots/thread
+  constexpr int BSL = (BN * (BK / 8)) / THREADS;   // B staging slots/thread
   bf16x8 areg[ASL], breg[BSL];
 
   // T14 software pipeline (guide §6 G15): issue chunk t+1's global loads,
@@ -153,8 +153,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 #pragma unroll
     for (int it = 0; it < ASL; ++it) {
       const int slot = it * THREADS + tid;
-      const int row = slot >> 3;
-      const int cb = (slot & 7) * 16;
+      const int row = slot / (BK / 8);
+      const int cb = (slot % (BK / 8)) * 16;
       bf16x8 v = {};
       if (RSCLIN) {
 #pragma unroll
@@ -191,8 +191,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 #pragma unroll
     for (int it = 0; it < BSL; ++it) {
       const int slot = it * THREADS + tid;
-      const int row = slot >> 3;
-      const int cb = (slot & 7) * 16;
+      const int row = slot / (BK / 8);
+      const int cb = (slot % (BK / 8)) * 16;
       const int k = k0 + row;
       bf16x8 v = {};
       if (k < a.K) {
@@ -239,14 +239,14 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 #pragma unroll
     for (int it = 0; it < ASL; ++it) {
       const int slot = it * THREADS + tid;
-      *reinterpret_cast<bf16x8*>(a_lds + swz(slot >> 3, (slot & 7) * 16)) =
-          areg[it];
+      *reinterpret_cast<bf16x8*>(
+          a_lds + swz<BK>(slot / (BK / 8), (slot % (BK / 8)) * 16)) = areg[it];
     }
 #pragma unroll
     for (int it = 0; it < BSL; ++it) {
       const int slot = it * THREADS + tid;
-      *reinterpret_cast<bf16x8*>(b_lds + swz(slot >> 3, (slot & 7) * 16)) =
-          breg[it];
+      *reinterpret_cast<bf16x8*>(
+          b_lds + swz<BK>(slot / (BK / 8), (slot % (BK / 8)) * 16)) = breg[it];
     }
   };
 
@@ -257,20 +257,20 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
   __syncthreads();
 
   for (int t = 0; t < nchunks; ++t) {
-    // ---- MFMA over the 64-wide chunk (2 x K=32 steps) ----
+    // ---- MFMA over the BK-wide chunk (BK/32 x K=32 steps) ----
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
+    for (int kk = 0; kk < BK / 32; ++kk) {
       const int cb = kk * 64 + ((lane >> 4) * 16);
       bf16x8 a_frag[FM], b_frag[FN];
 #pragma unroll
       for (int f = 0; f < FM; ++f) {
         const int arow = wm + f * 16 + (lane & 15);
-        a_frag[f] = *reinterpret_cast<const bf16x8*>(a_lds + swz(arow, cb));
+        a_frag[f] = *reinterpret_cast<const bf16x8*>(a_lds + swz<BK>(arow, cb));
       }
 #pragma unroll
       for (int f = 0; f < FN; ++f) {
         const int brow = wn + f * 16 + (lane & 15);
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(b_lds + swz(brow, cb));
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(b_lds + swz<BK>(brow, cb));
       }
 #pragma unroll
       for (int i = 0; i < FM; ++i)
@@ -326,7 +326,7 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
   }
 }
 
-template <int BM, int BN, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
+template <int BM, int BN, int BK, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
 void launch_one(const torch::Tensor& in, const torch::Tensor& w,
                 const c10::optional<torch::Tensor>& bias, torch::Tensor& out,
                 ConvArgs& a, int nz) {
@@ -335,7 +335,7 @@ void launch_one(const torch::Tensor& in, const torch::Tensor& w,
   dim3 grid(a.mblocks, ceil_div(a.K, BN), nz);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(
-      (conv2d_nhwc_fwd_kernel<BM, BN, KSIZE, STRIDE, FRAC, RSCLIN>), grid,
+      (conv2d_nhwc_fwd_kernel<BM, BN, BK, KSIZE, STRIDE, FRAC, RSCLIN>), grid,
       dim3(THREADS), 0, stream,
       reinterpret_cast<const __bf16*>(in.data_ptr()),
       reinterpret_cast<const __bf16*>(w.data_ptr()),
@@ -352,17 +352,23 @@ void dispatch_tile(const torch::Tensor& in, const torch::Tensor& w,
   const int M = a.Nb * a.HO * a.WO;
   const bool small = (long)ceil_div(M, 128) * ceil_div(a.K, 128) < 160;
   const bool narrow = a.K <= 64;  // half a BN=128 tile would be masked out
+  const bool deep = a.C >= 128;   // BK=128 halves the barrier count
   if (rsclin) {
     if (narrow)
-      launch_one<128, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+      launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
     else
-      launch_one<128, 128, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+      launch_one<128, 128, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
   } else if (small) {
-    launch_one<64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    launch_one<64, 64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   } else if (narrow) {
-    launch_one<128, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    if (deep)
+      launch_one<128, 64, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    else
+      launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+  } else if (deep) {
+    launch_one<128, 128, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   } else {
-    launch_one<128, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    launch_one<128, 128, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
   }
 }
 
