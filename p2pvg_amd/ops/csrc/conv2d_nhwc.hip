@@ -352,7 +352,10 @@ void dispatch_tile(const torch::Tensor& in, const torch::Tensor& w,
   const int M = a.Nb * a.HO * a.WO;
   const bool small = (long)ceil_div(M, 128) * ceil_div(a.K, 128) < 160;
   const bool narrow = a.K <= 64;  // half a BN=128 tile would be masked out
-  const bool deep = a.C >= 128;   // BK=128 halves the barrier count
+  // BK=128 was measured 20-60% SLOWER on the deep-C shapes (64 KB LDS cuts
+  // occupancy 4->2 blocks/CU; the barrier savings don't cover it — matches
+  // the guide's BK=128 regression note). Keep BK=64 everywhere.
+  const bool deep = false;
   if (rsclin) {
     if (narrow)
       launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
