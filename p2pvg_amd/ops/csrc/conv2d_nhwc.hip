@@ -71,12 +71,10 @@ template <int BM, int BN, int BK, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
 __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     const __bf16* __restrict__ in, const __bf16* __restrict__ w,
     const float* __restrict__ bias, __bf16* __restrict__ out, ConvArgs a) {
-  // double-buffered tiles: the register file (196 regs/wave) caps occupancy
-  // at 2 blocks/CU regardless, so 2x(BM+BN)*BK*2 bytes of LDS are free and
-  // buy the one-barrier-per-chunk schedule (guide G15 write-after-barrier)
-  constexpr int BUFB = (BM + BN) * (BK * 2);
-  __shared__ __align__(16) char lds[2 * BUFB + BM * 16];
-  long* pix_out = reinterpret_cast<long*>(lds + 2 * BUFB);
+  __shared__ __align__(16) char lds[(BM + BN) * (BK * 2) + BM * 16];
+  char* a_lds = lds;
+  char* b_lds = lds + BM * (BK * 2);
+  long* pix_out = reinterpret_cast<long*>(lds + (BM + BN) * (BK * 2));
   int* pix_off = reinterpret_cast<int*>(pix_out + BM);
   short* pix_hi = reinterpret_cast<short*>(pix_off + BM);
   short* pix_wi = pix_hi + BM;
@@ -237,9 +235,7 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     }
   };
 
-  auto write_chunk = [&](int buf) {
-    char* a_lds = lds + buf * BUFB;
-    char* b_lds = a_lds + BM * (BK * 2);
+  auto write_chunk = [&]() {
 #pragma unroll
     for (int it = 0; it < ASL; ++it) {
       const int slot = it * THREADS + tid;
@@ -254,19 +250,13 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     }
   };
 
-  // prologue: chunk 0 into buffer 0, chunk 1 in flight
+  // prologue: chunk 0 into LDS, chunk 1 in flight
   load_chunk(0);
-  write_chunk(0);
+  write_chunk();
   if (nchunks > 1) load_chunk(1);
   __syncthreads();
 
   for (int t = 0; t < nchunks; ++t) {
-    // write t+1 into the other buffer (regs loaded last iteration), re-issue
-    // t+2 immediately, then compute t — one barrier per chunk
-    if (t + 1 < nchunks) write_chunk((t + 1) & 1);
-    if (t + 2 < nchunks) load_chunk(t + 2);
-    char* a_lds = lds + (t & 1) * BUFB;
-    char* b_lds = a_lds + BM * (BK * 2);
     // ---- MFMA over the BK-wide chunk (BK/32 x K=32 steps) ----
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
@@ -289,7 +279,12 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    if (t + 1 < nchunks) {
+      __syncthreads();          // everyone done reading chunk t
+      write_chunk();            // chunk t+1 registers -> LDS
+      if (t + 2 < nchunks) load_chunk(t + 2);  // re-issue immediately
+      __syncthreads();          // chunk t+1 visible
+    }
   }
 
   // ---- epilogue ----
