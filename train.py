@@ -78,6 +78,13 @@ def main():
     args = parser.parse_args()
     cfg = config_from_args(args)
 
+    if cfg.deterministic:
+        # CI/debug mode: the hand-written kernels use split-K
+        # atomics whose accumulation order is nondeterministic; route hot ops
+        # through the ATen path and enable torch's deterministic algorithms.
+        os.environ["P2PVG_KERNELS"] = "torch"
+        torch.use_deterministic_algorithms(True, warn_only=True)
+
     rank, world = setup_distributed(cfg)
     is_main = rank == 0
 
