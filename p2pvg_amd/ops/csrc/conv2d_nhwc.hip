@@ -131,7 +131,6 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const bool cvec = (a.C % 8) == 0;
-  const bool full = !RSCLIN && (a.C % BK) == 0;  // wave-uniform
   const int RSC = KSIZE * KSIZE * a.C;
   // chunk loop bounds
   const int n_outer = RSCLIN ? 1 : KSIZE * KSIZE;
@@ -204,10 +203,6 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
               v[j] = in[(long)pix_off[row] + ((long)hi * a.W + wi) * a.C + c];
           }
         }
-      } else if (full) {
-        // C % BK == 0: every chunk element is in range
-        if (abase[it] >= 0)
-          v = *reinterpret_cast<const bf16x8*>(in + abase[it] + c0 + (cb >> 1));
       } else {
         const int c = c0 + (cb >> 1);
         if (abase[it] >= 0 && c < a.C) {
@@ -250,9 +245,6 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
             }
           }
         }
-      } else if (full) {
-        if (bbase[it] >= 0)
-          v = *reinterpret_cast<const bf16x8*>(w + bbase[it] + c0 + (cb >> 1));
       } else {
         const int c = c0 + (cb >> 1);
         if (bbase[it] >= 0 && c < a.C) {
