@@ -227,159 +227,6 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
   }
 }
 
-
-// ---------------------------------------------------------------------------
-// k3/stride-1 specialization: one block owns a (32 x 32) channel tile and a
-// slab of image planes; for each output row it stages the Y row and keeps a
-// 3-row ring of X rows in LDS, so ALL NINE taps are computed from data read
-// from HBM exactly once (the generic kernel above re-reads both tensors once
-// per tap — 9x the traffic). The s-axis shifts are register selects over
-// adjacent hardware-transpose reads; edge pixels are masked statically.
-// Requires W % 8 == 0 (pixel runs align with rows); rows are zero-padded to
-// the 32-pixel MFMA K-step.
-// ---------------------------------------------------------------------------
-
-template <int DUMMY>
-__global__ __launch_bounds__(THREADS) void conv2d_wgrad_k3s1_kernel(
-    const __bf16* __restrict__ Y,  // (N, H, W, B) — same spatial dims as X
-    const __bf16* __restrict__ X,  // (N, H, W, A)
-    float* __restrict__ ws,        // (B, 3, 3, A) fp32, pre-zeroed
-    int Nb, int H, int W, int B, int A, int planes_per_block) {
-  // LDS: Y row [Wp px][32 ch] subtiled + 3-slot X row ring, Wp = padded W
-  const int Wp = (W + 31) & ~31;
-  const int rowb = Wp * 32 * 2;              // bytes per staged row
-  extern __shared__ __align__(16) char lds[];
-  char* yrow = lds;                          // [Wp/4][2][4][16] subtiles
-  char* xring = lds + rowb;                  // 3 x rowb
-
-  const int at_blocks = (A + 31) / 32;
-  const int b0 = ((int)blockIdx.x / at_blocks) * 32;
-  const int a0 = ((int)blockIdx.x % at_blocks) * 32;
-  const int n_begin = (int)blockIdx.y * planes_per_block;
-  const int n_end = min(n_begin + planes_per_block, Nb);
-
-  const int tid = (int)threadIdx.x;
-  const int wid = tid >> 6;
-  const int lane = tid & 63;
-  const int wm = (wid >> 1) * 16;            // wave row (b) base in tile
-  const int wn = (wid & 1) * 16;             // wave col (a) base in tile
-
-  // 9 accumulators (tap-major), one 16x16 fragment each
-  f32x4 acc[9];
-#pragma unroll
-  for (int t = 0; t < 9; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
-
-  // stage one row of a tensor into a subtiled LDS image
-  // thread t covers pixel px = t/4, channel chunk (t%4)*8 (32ch = 4 chunks)
-  auto stage_row = [&](char* dst, const __bf16* src_base, int c0, int C,
-                       int n, int row) {
-    const int px = tid >> 2;
-    const int ch0 = (tid & 3) * 8;
-    if (px < Wp) {
-      bf16x8 v = {};
-      if (px < W && c0 + ch0 < C) {
-        const __bf16* src =
-            src_base + (((long)n * H + row) * W + px) * C + c0 + ch0;
-        if (c0 + ch0 + 8 <= C) {
-          v = *reinterpret_cast<const bf16x8*>(src);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            if (c0 + ch0 + j < C) v[j] = src[j];
-        }
-      }
-      // subtile layout [px/4][ch/16][4][16]
-      *reinterpret_cast<bf16x8*>(dst + ((px >> 2) * 2 + (ch0 >> 4)) * 128 +
-                                 (px & 3) * 32 + (ch0 & 15) * 2) = v;
-    }
-  };
-
-  for (int n = n_begin; n < n_end; ++n) {
-    // prologue: X rows 0 and 1 into ring slots 0, 1
-    stage_row(xring + 0 * rowb, X, a0, A, n, 0);
-    if (H > 1) stage_row(xring + 1 * rowb, X, a0, A, n, 1);
-
-    for (int ho = 0; ho < H; ++ho) {
-      // stage Y[ho] and the ring row ho+1
-      stage_row(yrow, Y, b0, B, n, ho);
-      if (ho + 1 < H && ho > 0)
-        stage_row(xring + ((ho + 1) % 3) * rowb, X, a0, A, n, ho + 1);
-      __syncthreads();
-
-      for (int kk = 0; kk < Wp / 32; ++kk) {
-        const int pe = kk * 32 + (lane >> 4) * 8;  // pixel base, 8-aligned
-        // Y fragment: channels b (wm + lane&15), pixels pe..pe+7
-        u16x4 ylo, yhi;
-        tr16_issue(yrow, wm + (lane & 15), pe, lane, 2, ylo, yhi);
-        // X pairs at pe-8, pe, pe+8 per valid x-row
-        u16x4 xl[3][2], xc[3][2], xr[3][2];
-#pragma unroll
-        for (int r = 0; r < 3; ++r) {
-          const int xrow = ho + r - 1;
-          if (xrow < 0 || xrow >= H) continue;
-          char* xt = xring + (((xrow % 3) + 3) % 3) * rowb;
-          const int ch = wn + (lane & 15);
-          if (pe >= 8) tr16_issue(xt, ch, pe - 8, lane, 2, xl[r][0], xl[r][1]);
-          tr16_issue(xt, ch, pe, lane, 2, xc[r][0], xc[r][1]);
-          if (pe + 8 < Wp) tr16_issue(xt, ch, pe + 8, lane, 2, xr[r][0], xr[r][1]);
-        }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-        __builtin_amdgcn_sched_barrier(0);
-
-        const bf16x8 a_frag = tr16_combine(ylo, yhi);
-#pragma unroll
-        for (int r = 0; r < 3; ++r) {
-          const int xrow = ho + r - 1;
-          if (xrow < 0 || xrow >= H) continue;
-          typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
-          u16x8 c8 = __builtin_shufflevector(xc[r][0], xc[r][1], 0, 1, 2, 3,
-                                             4, 5, 6, 7);
-          u16x8 l8 = pe >= 8 ? __builtin_shufflevector(xl[r][0], xl[r][1], 0,
-                                                       1, 2, 3, 4, 5, 6, 7)
-                             : (u16x8){0, 0, 0, 0, 0, 0, 0, 0};
-          u16x8 r8 = pe + 8 < Wp
-                         ? __builtin_shufflevector(xr[r][0], xr[r][1], 0, 1,
-                                                   2, 3, 4, 5, 6, 7)
-                         : (u16x8){0, 0, 0, 0, 0, 0, 0, 0};
-          // s = 1 (center): staged pixels pe..pe+7
-          bf16x8 f1 = __builtin_bit_cast(bf16x8, c8);
-          // s = 0: pixels pe-1..pe+6 (element 0 from the left pair)
-          u16x8 s0 = __builtin_shufflevector(l8, c8, 7, 8, 9, 10, 11, 12, 13,
-                                             14);
-          if (pe == 0) s0[0] = 0;  // wo-1 < 0 pad
-          bf16x8 f0 = __builtin_bit_cast(bf16x8, s0);
-          // s = 2: pixels pe+1..pe+8
-          u16x8 s2 = __builtin_shufflevector(c8, r8, 1, 2, 3, 4, 5, 6, 7, 8);
-          if (pe + 8 >= W) s2[7] = 0;  // wo+1 >= W pad (also kills row pad)
-          bf16x8 f2 = __builtin_bit_cast(bf16x8, s2);
-
-          acc[r * 3 + 0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, f0, acc[r * 3 + 0], 0, 0, 0);
-          acc[r * 3 + 1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, f1, acc[r * 3 + 1], 0, 0, 0);
-          acc[r * 3 + 2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag, f2, acc[r * 3 + 2], 0, 0, 0);
-        }
-      }
-      __syncthreads();
-    }
-  }
-
-  // flush: dW[b, r, s, a] += acc
-#pragma unroll
-  for (int t = 0; t < 9; ++t) {
-    const int r = t / 3, sx = t % 3;
-    const int a = a0 + wn + (lane & 15);
-    if (a >= A) continue;
-#pragma unroll
-    for (int v = 0; v < 4; ++v) {
-      const int b = b0 + wm + (lane >> 4) * 4 + v;
-      if (b < B)
-        atomicAdd(&ws[(((long)b * 3 + r) * 3 + sx) * A + a], acc[t][v]);
-    }
-  }
-}
-
 }  // namespace
 
 // Y: channels_last (N,B,HO,WO) bf16; X: channels_last (N,A,H,W) bf16.
@@ -398,26 +245,6 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
 
   auto ws = torch::zeros({B, (long)R, (long)S, A},
                          Y.options().dtype(torch::kFloat32));
-
-  // k3/s1 with row-aligned pixel runs: the tap-dedup kernel reads each
-  // tensor once instead of 9 times
-  if (R == 3 && S == 3 && stride == 1 && pad == 1 && W % 8 == 0 && W >= 32 &&
-      HO == H && WO == W && splitp <= 0) {
-    const int bt = ceil_div(B, 32), at = ceil_div(A, 32);
-    // cap the per-address atomic contention (= planes per channel-pair) at 64
-    int ppb = ceil_div(Nb, 64);
-    while ((long)bt * at * ceil_div(Nb, ppb) > 1024 && ppb < Nb) ppb *= 2;
-    const int Wp = (W + 31) & ~31;
-    const size_t shmem = (size_t)(Wp * 32 * 2) * 4;  // Y row + 3-row X ring
-    dim3 grid(bt * at, ceil_div(Nb, ppb));
-    auto stream = at::cuda::getCurrentCUDAStream();
-    hipLaunchKernelGGL((conv2d_wgrad_k3s1_kernel<0>), grid, dim3(THREADS),
-                       shmem, stream,
-                       reinterpret_cast<const __bf16*>(Y.data_ptr()),
-                       reinterpret_cast<const __bf16*>(X.data_ptr()),
-                       ws.data_ptr<float>(), Nb, H, W, B, A, ppb);
-    return ws;
-  }
 
   const int BTB = B >= 128 ? 128 : 64;
   const int BTA = A >= 128 ? 128 : 64;
