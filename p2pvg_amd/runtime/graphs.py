@@ -1,11 +1,9 @@
 """hipGraph capture of the whole training step (SURVEY §5.7 MI355X plan).
 
 The per-timestep recurrence launches thousands of small kernels per training
-step; at the measured batch sizes the GPU sits idle well over half the wall
-time between launches (profiles/ shows the kernel-trace evidence). Capturing
-the entire step — forward over all processed timesteps, both backward phases,
-and all five (capturable) Adam steps — into one hipGraph and replaying it
-removes every launch gap.
+step; capturing the entire step — forward over all processed timesteps, both
+backward phases, and all five (capturable) Adam steps — into one hipGraph and
+replaying it removes every launch gap and all host-side dispatch.
 
 Dynamic control flow is handled exactly as SURVEY §5.7 prescribes:
 - The skip gate and dynamic sequence length are HOST-side (plan_step); the
@@ -17,14 +15,22 @@ Dynamic control flow is handled exactly as SURVEY §5.7 prescribes:
 - Reparameterization randomness uses the device philox generator, which
   torch.cuda.CUDAGraph captures/advances correctly.
 
+Memory-pool discipline: every graph gets its OWN memory pool. Sharing one
+pool across independently-replayed graphs corrupts state when replays happen
+out of capture order (observed as negative MSE in long dynamic-length runs).
+Since each pool pins a full step's activations, the number of live graphs is
+capped (`max_graphs`); rarer (seq_len, skip-pattern) keys run eagerly through
+the same model code path.
+
 Capture warmup runs the step function a few times on a side stream (the
-standard recipe: materializes grads, Adam state, MIOpen solutions, autocast
-weight casts) — those warmup iterations ARE real optimizer steps; for
-benchmarking they land in the warmup phase, for training they are ordinary
-extra steps on the first batch of that shape.
+standard recipe: materializes grads, Adam state, autocast weight casts) —
+those warmup iterations ARE real optimizer steps; for benchmarking they land
+in the warmup phase, for training they are ordinary extra steps on the first
+batch of that shape.
 """
 from __future__ import annotations
 
+import contextlib
 from typing import Dict, Optional, Tuple
 
 import torch
@@ -38,23 +44,24 @@ class _GraphEntry:
 
 class GraphedTrainStep:
     def __init__(self, model, amp_dtype: Optional[torch.dtype] = None,
-                 warmup_iters: int = 3):
+                 warmup_iters: int = 3, max_graphs: int = 16):
         self.model = model
         self.amp_dtype = amp_dtype
         self.warmup_iters = warmup_iters
+        self.max_graphs = max_graphs
         self.graphs: Dict[Tuple, _GraphEntry] = {}
-        self.pool = None  # shared memory pool across all captured graphs
+        self.eager_keys = set()
+
+    def _amp_ctx(self, cache_enabled: bool):
+        if self.amp_dtype is not None:
+            return torch.autocast("cuda", dtype=self.amp_dtype,
+                                  cache_enabled=cache_enabled)
+        return contextlib.nullcontext()
 
     def _inner(self, entry: _GraphEntry, plan: StepPlan):
         model = self.model
         model.zero_grad(set_to_none=False)
-        if self.amp_dtype is not None:
-            ctx = torch.autocast("cuda", dtype=self.amp_dtype, cache_enabled=False)
-        else:
-            import contextlib
-
-            ctx = contextlib.nullcontext()
-        with ctx:
+        with self._amp_ctx(cache_enabled=False):
             losses = model._compute_losses(
                 entry.prev_buf, entry.cur_buf, entry.tun, entry.dts, plan
             )
@@ -63,6 +70,14 @@ class GraphedTrainStep:
         entry.loss_out.copy_(
             torch.stack([mse.detach(), kld.detach(), cpc.detach(), align.detach()])
         )
+
+    def _eager_step(self, plan, prev, cur, tun, dts):
+        model = self.model
+        model.zero_grad(set_to_none=False)
+        with self._amp_ctx(cache_enabled=True):
+            losses = model._compute_losses(prev, cur, tun, dts, plan)
+        model._backward_and_step(*losses)
+        return tuple(v.detach() for v in losses)
 
     def _capture(self, plan: StepPlan, prev: torch.Tensor, cur: torch.Tensor,
                  tun: torch.Tensor, dts: torch.Tensor) -> _GraphEntry:
@@ -83,13 +98,9 @@ class GraphedTrainStep:
         torch.cuda.synchronize()
 
         g = torch.cuda.CUDAGraph()
-        if self.pool is None:
-            with torch.cuda.graph(g):
-                self._inner(entry, plan)
-            self.pool = g.pool()
-        else:
-            with torch.cuda.graph(g, pool=self.pool):
-                self._inner(entry, plan)
+        # private pool per graph: see the module docstring
+        with torch.cuda.graph(g):
+            self._inner(entry, plan)
         entry.graph = g
         return entry
 
@@ -111,6 +122,11 @@ class GraphedTrainStep:
         key = plan.graph_key
         entry = self.graphs.get(key)
         if entry is None:
+            if key in self.eager_keys or len(self.graphs) >= self.max_graphs:
+                self.eager_keys.add(key)
+                losses = self._eager_step(plan, prev, cur, tun, dts)
+                inv = 1.0 / seq_len
+                return tuple(v * inv for v in losses)
             # capture records but does not execute; fall through to replay
             entry = self._capture(plan, prev, cur, tun, dts)
             self.graphs[key] = entry
