@@ -53,6 +53,7 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor gout, torch::Tensor idx, long H,
                              long W);
 torch::Tensor upsample2x_fwd(torch::Tensor in);
 torch::Tensor upsample2x_bwd(torch::Tensor gout);
+torch::Tensor channel_sum_nhwc(torch::Tensor x);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
@@ -71,6 +72,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd_train", &bn_act_fwd_train, "fused BN+act train fwd (gfx950)");
   m.def("bn_act_fwd_eval", &bn_act_fwd_eval, "fused BN+act eval fwd (gfx950)");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act bwd (gfx950)");
+  m.def("channel_sum_nhwc", &channel_sum_nhwc,
+        "NHWC per-channel sum, fp32 out (gfx950)");
   m.def("conv2d_nhwc_wgrad", &conv2d_nhwc_wgrad,
         "NHWC wgrad, split-K over pixels, fp32 workspace (gfx950 MFMA)");
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC 2x2/s2 maxpool fwd (gfx950)");

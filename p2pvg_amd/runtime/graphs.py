@@ -96,6 +96,11 @@ class GraphedTrainStep:
                 self._inner(entry, plan)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
+        # release warmup's cached segments: expandable_segments is a no-op on
+        # this ROCm build, and the graph's private pool cannot reuse the
+        # allocator's fragmented free blocks (observed: 116 GB reserved but
+        # unallocated OOM'ing a vgg_128 batch-128 capture)
+        torch.cuda.empty_cache()
 
         g = torch.cuda.CUDAGraph()
         # private pool per graph: see the module docstring

@@ -179,3 +179,15 @@ def test_model_step_bf16_autocast():
         losses = model(x, 0, 5)
     torch.cuda.synchronize()
     assert all(torch.isfinite(v) for v in losses)
+
+
+@pytest.mark.parametrize("N,C,H,W", [(4, 64, 32, 32), (3, 8, 5, 7),
+                                     (2, 512, 4, 4), (7, 128, 16, 16)])
+def test_channel_sum_nhwc(ext, N, C, H, W):
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    got = ext.channel_sum_nhwc(x)
+    want = x.float().sum(dim=(0, 2, 3))
+    assert got.dtype == torch.float32
+    torch.testing.assert_close(got, want, rtol=1e-3, atol=1e-2)
