@@ -126,15 +126,29 @@ class GraphedTrainStep:
 
         key = plan.graph_key
         entry = self.graphs.get(key)
+        if (entry is None and key not in self.eager_keys
+                and len(self.graphs) < self.max_graphs):
+            # capture records but does not execute; falls through to replay
+            try:
+                entry = self._capture(plan, prev, cur, tun, dts)
+                self.graphs[key] = entry
+            except torch.OutOfMemoryError:
+                # capture needs headroom beyond the eager peak (side-stream
+                # warmup segments + the graph's private pool cannot share
+                # the allocator's fragmented free blocks, and
+                # expandable_segments is a no-op on this ROCm build).
+                # Large-batch configs train eagerly instead of not at all.
+                # The fallback itself runs OUTSIDE this handler: the
+                # exception's traceback pins every warmup frame (and so all
+                # of that step's activations) until the handler exits.
+                entry = None
         if entry is None:
-            if key in self.eager_keys or len(self.graphs) >= self.max_graphs:
+            if key not in self.eager_keys:
                 self.eager_keys.add(key)
-                losses = self._eager_step(plan, prev, cur, tun, dts)
-                inv = 1.0 / seq_len
-                return tuple(v * inv for v in losses)
-            # capture records but does not execute; fall through to replay
-            entry = self._capture(plan, prev, cur, tun, dts)
-            self.graphs[key] = entry
+                torch.cuda.empty_cache()
+            losses = self._eager_step(plan, prev, cur, tun, dts)
+            inv = 1.0 / seq_len
+            return tuple(v * inv for v in losses)
 
         entry.prev_buf.copy_(prev)
         entry.cur_buf.copy_(cur)
