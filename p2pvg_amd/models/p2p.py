@@ -249,9 +249,9 @@ class P2PModel(nn.Module):
                     torch.cat([h, zt_p, time_until_cp, delta_time], 1)
                 )
                 x_pred_p = self.decoder([h_pred_p, skip])
-                cpc_loss = self.mse_criterion(x_pred_p, x_cp)
+                cpc_loss = ops.frame_mse(x_pred_p, x_cp)
 
-            mse_loss = mse_loss + self.mse_criterion(x_pred, cur_frames[k])
+            mse_loss = mse_loss + ops.frame_mse(x_pred, cur_frames[k])
             kld_loss = kld_loss + self.kl_criterion(mu, logvar, mu_p, logvar_p)
 
             h_prev = h

@@ -47,6 +47,14 @@ def hip_available() -> bool:
     return _load_hip_ext() is not None
 
 
+def frame_mse(a, b):
+    """MSE over frame tensors: fused one-pass reduction on the HIP path
+    (large bf16 tensors), F.mse_loss otherwise."""
+    from .losses import fused_mse
+
+    return fused_mse(a, b)
+
+
 def _want_hip(t: torch.Tensor) -> bool:
     mode = backend_mode()
     if mode == "torch":

@@ -78,7 +78,7 @@ def _channel_sum(gout):
     if (gout.is_cuda and gout.dtype == torch.bfloat16
             and gout.is_contiguous(memory_format=CL)):
         C = gout.shape[1]
-        if C % 8 == 0 and 256 % (C // 8) == 0:
+        if C <= 4 or (C % 8 == 0 and 256 % (C // 8) == 0):
             return _ext().channel_sum_nhwc(gout)
     return gout.sum(dim=(0, 2, 3), dtype=torch.float32)
 

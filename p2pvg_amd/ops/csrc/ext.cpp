@@ -54,6 +54,7 @@ torch::Tensor maxpool2x2_bwd(torch::Tensor gout, torch::Tensor idx, long H,
 torch::Tensor upsample2x_fwd(torch::Tensor in);
 torch::Tensor upsample2x_bwd(torch::Tensor gout);
 torch::Tensor channel_sum_nhwc(torch::Tensor x);
+torch::Tensor sqdiff_sum(torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
@@ -86,4 +87,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_adam", &multi_tensor_adam, "multi-tensor Adam (gfx950)");
   m.def("gaussian_kl_fwd", &gaussian_kl_fwd, "fused gaussian KL fwd (gfx950)");
   m.def("gaussian_kl_bwd", &gaussian_kl_bwd, "fused gaussian KL bwd (gfx950)");
+  m.def("sqdiff_sum", &sqdiff_sum, "fused sum((a-b)^2), bf16/f32 in fp32 out");
 }

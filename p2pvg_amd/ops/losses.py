@@ -29,3 +29,44 @@ class GaussianKLFn(torch.autograd.Function):
             mu1, lv1, mu2, lv2, dout.contiguous(), ctx.denom
         )
         return dmu1, dlv1, dmu2, dlv2, None
+
+
+class FusedMSEFn(torch.autograd.Function):
+    """mean((a-b)^2) with the one-pass fp32-accumulating reduction kernel
+    (csrc/gaussian_kl.hip sqdiff_sum). ATen's mse_loss on
+    the bf16 frame tensors dispatches a ~320us ReduceOp per call (~25x off
+    bandwidth) plus cast passes; this reads each tensor once."""
+
+    @staticmethod
+    def forward(ctx, a, b):
+        out = _ext().sqdiff_sum(a, b) / a.numel()
+        ctx.save_for_backward(a, b)
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        a, b = ctx.saved_tensors
+        ga = gb = None
+        # d/da mean((a-b)^2) = 2/N * (a-b) * gout
+        d = (a - b if a.dtype == b.dtype else a.float() - b) * (
+            gout * (2.0 / a.numel())
+        )
+        if ctx.needs_input_grad[0]:
+            ga = d.to(a.dtype) if d.dtype != a.dtype else d
+        if ctx.needs_input_grad[1]:
+            gb = (-d).to(b.dtype)
+        return ga, gb
+
+
+def fused_mse(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """MSE that routes large same-layout frame tensors through the fused
+    reduction kernel; everything else falls back to F.mse_loss."""
+    from . import _want_hip
+
+    if (_want_hip(a) and a.dtype == torch.bfloat16
+            and b.dtype in (torch.bfloat16, torch.float32)
+            and a.shape == b.shape and a.stride() == b.stride()
+            and a.numel() % 8 == 0 and a.numel() >= 32768
+            and not a.is_sparse and not b.is_sparse):
+        return FusedMSEFn.apply(a, b)
+    return torch.nn.functional.mse_loss(a, b)

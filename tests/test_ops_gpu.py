@@ -182,7 +182,8 @@ def test_model_step_bf16_autocast():
 
 
 @pytest.mark.parametrize("N,C,H,W", [(4, 64, 32, 32), (3, 8, 5, 7),
-                                     (2, 512, 4, 4), (7, 128, 16, 16)])
+                                     (2, 512, 4, 4), (7, 128, 16, 16),
+                                     (16, 3, 64, 64), (5, 1, 32, 32)])
 def test_channel_sum_nhwc(ext, N, C, H, W):
     torch.manual_seed(0)
     x = torch.randn(N, C, H, W, device="cuda").to(torch.bfloat16) \
@@ -191,3 +192,37 @@ def test_channel_sum_nhwc(ext, N, C, H, W):
     want = x.float().sum(dim=(0, 2, 3))
     assert got.dtype == torch.float32
     torch.testing.assert_close(got, want, rtol=1e-3, atol=1e-2)
+
+
+@pytest.mark.parametrize("b_dtype", [torch.bfloat16, torch.float32])
+def test_fused_mse_matches_fp32(ext, b_dtype):
+    from p2pvg_amd.ops.losses import FusedMSEFn
+
+    torch.manual_seed(0)
+    a = torch.rand(8, 3, 64, 64, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    b = torch.rand(8, 3, 64, 64, device="cuda").to(b_dtype) \
+        .contiguous(memory_format=torch.channels_last)
+    out = FusedMSEFn.apply(a, b)
+    ref = torch.nn.functional.mse_loss(a.detach().float(), b.float())
+    torch.testing.assert_close(out, ref, rtol=1e-3, atol=1e-5)
+
+    out.backward()
+    a2 = a.detach().clone().float().requires_grad_(True)
+    torch.nn.functional.mse_loss(a2, b.float()).backward()
+    torch.testing.assert_close(a.grad.float(), a2.grad,
+                               rtol=2e-2, atol=2e-3)
+
+
+def test_fused_mse_routing_in_model_losses():
+    """frame_mse must route the big frame tensors through the kernel on GPU
+    and return the same value as F.mse_loss."""
+    from p2pvg_amd import ops
+
+    a = torch.rand(64, 3, 64, 64, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    b = torch.rand(64, 3, 64, 64, device="cuda") \
+        .contiguous(memory_format=torch.channels_last)
+    got = ops.frame_mse(a, b)
+    ref = torch.nn.functional.mse_loss(a.float(), b)
+    torch.testing.assert_close(got, ref, rtol=1e-3, atol=1e-5)
