@@ -187,6 +187,12 @@ class P2PModel(nn.Module):
         hipGraph-capturable with the frames/scalars as graph inputs.
         """
         cfg = self.cfg
+        if prev_frames.is_cuda:
+            # one cast+flip pass per step instead of per conv use; recorded
+            # inside the hipGraph so replays refresh too
+            from ..ops.conv import refresh_conv_shadows
+
+            refresh_conv_shadows(self)
         n = len(plan.proc)
         batch_size = prev_frames[0].shape[0]
         device = prev_frames.device
@@ -381,6 +387,10 @@ class P2PModel(nn.Module):
 
         batch_size = x[0].shape[0]
         device = x[0].device
+        if x[0].is_cuda:
+            from ..ops.conv import refresh_conv_shadows
+
+            refresh_conv_shadows(self)
 
         gen_seq = [x[0]]
         x_in = x[0]

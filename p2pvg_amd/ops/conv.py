@@ -83,37 +83,106 @@ def _channel_sum(gout):
     return gout.sum(dim=(0, 2, 3), dtype=torch.float32)
 
 
+@torch.no_grad()
+def _fill_conv_shadow(conv, sh):
+    w = conv.weight
+    if isinstance(conv, nn.ConvTranspose2d):
+        sh["b"].copy_(w)  # plain (Ci,Co,k,k)
+        if conv.stride[0] == 1:
+            sh["f"].copy_(w.flip(2, 3).transpose(0, 1))
+        else:
+            sh["f"].copy_(w.transpose(0, 1))
+    else:
+        sh["f"].copy_(w)  # plain (K,C,k,k)
+        if conv.stride[0] == 1:
+            sh["b"].copy_(w.flip(2, 3).transpose(0, 1))
+        else:
+            sh["b"].copy_(w.transpose(0, 1))
+    sh["ver"] = conv.weight._version
+
+
+def _conv_shadows(conv):
+    """Per-step bf16 weight shadows: 'f' = forward operand, 'b' = backward
+    (dgrad / gemm) operand, both channels_last bf16 and refreshed ONCE per
+    weight version instead of cast+flip+transposed at every one of the ~29
+    timestep uses. refresh_conv_shadows() force-refreshes (needed inside a
+    hipGraph capture, where replays never re-run this Python)."""
+    sh = getattr(conv, "_p2pvg_shadow", None)
+    if sh is None:
+        w = conv.weight
+        a, bdim, k, _ = w.shape
+        if isinstance(conv, nn.ConvTranspose2d):
+            fshape, bshape = (bdim, a, k, k), (a, bdim, k, k)
+        else:
+            fshape, bshape = (a, bdim, k, k), (bdim, a, k, k)
+        mk = lambda s: torch.empty(  # noqa: E731
+            s, device=w.device, dtype=torch.bfloat16, memory_format=CL
+        )
+        conv._p2pvg_shadow = sh = {"f": mk(fshape), "b": mk(bshape), "ver": -1}
+    if sh["ver"] != conv.weight._version:
+        _fill_conv_shadow(conv, sh)
+    return sh
+
+
+def refresh_conv_shadows(module):
+    """Force-refresh every conv shadow under `module`. Call once per training
+    step BEFORE the forward (recorded inside the hipGraph, so replays refresh
+    too) and at the start of generation."""
+    from . import backend_mode, hip_available
+
+    if backend_mode() == "torch" or not hip_available():
+        return
+    for m in module.modules():
+        if (isinstance(m, (nn.Conv2d, nn.ConvTranspose2d))
+                and (isinstance(m, (Conv2d, ConvTranspose2d))
+                     or hasattr(m, "_p2pvg_shadow"))
+                and m.weight.is_cuda):
+            sh = _conv_shadows(m)
+            _fill_conv_shadow(m, sh)
+
+
 class Conv2dNHWCFn(torch.autograd.Function):
     """y = act(conv2d(x, w, b, stride, pad)); optional per-channel sum/sumsq
     stats of the output (for the fused BatchNorm). Returns (y, stats)."""
 
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int = 0,
-                want_stats: bool = False):
+                want_stats: bool = False, w_fwd=None, w_bwd=None):
+        # w receives the weight gradient. With w_fwd/w_bwd (per-step bf16
+        # shadows, see refresh_conv_shadows), w is the raw fp32 param and
+        # the per-use cast / flip / transpose kernels disappear; without
+        # them w must already be bf16 channels_last (legacy direct calls).
         ext = _ext()
-        k = w.shape[2]
+        wf = w_fwd if w_fwd is not None else w
+        k = wf.shape[2]
         b32 = b.float() if b is not None else None
         stats = (
-            torch.zeros(64, 2, w.shape[0], device=x.device, dtype=torch.float32)
+            torch.zeros(64, 2, wf.shape[0], device=x.device, dtype=torch.float32)
             if want_stats else None
         )
         # degenerate whole-image conv (k == H, pad 0): plain GEMM
         gemm = pad == 0 and k == x.shape[2] and k == x.shape[3]
         if gemm:
-            out = torch.mm(_nhwc_flat(x), _nhwc_flat(w).t())
+            out = torch.mm(_nhwc_flat(x), _nhwc_flat(wf).t())
             if b is not None:
                 out = out + b.to(out.dtype)
             out = _act_fwd_torch(out, act)
             if want_stats:
                 stats = _gemm_stats(out).unsqueeze(0)
-            out = out.view(x.shape[0], w.shape[0], 1, 1).contiguous(
+            out = out.view(x.shape[0], wf.shape[0], 1, 1).contiguous(
                 memory_format=CL
             )
+            ctx.save_for_backward(x, wf, out if act != 0 else None)
+            ctx.shadow_bwd = False
         else:
-            out = ext.conv2d_nhwc_fwd(x, w, b32, stride, pad, act, stats)
-        ctx.save_for_backward(x, w, out if act != 0 else None)
+            out = ext.conv2d_nhwc_fwd(x, wf, b32, stride, pad, act, stats)
+            wb = w_bwd if w_bwd is not None else wf
+            ctx.save_for_backward(x, wb, out if act != 0 else None)
+            ctx.shadow_bwd = w_bwd is not None
+        ctx.with_shadows = w_fwd is not None or w_bwd is not None
         ctx.stride, ctx.pad, ctx.has_bias, ctx.gemm = stride, pad, b is not None, gemm
         ctx.act = act
+        ctx.wdtype = w.dtype
         if stats is None:
             stats = torch.empty(0, device=x.device)
         ctx.mark_non_differentiable(stats)
@@ -142,26 +211,38 @@ class Conv2dNHWCFn(torch.autograd.Function):
                 dw = torch.mm(g2.t(), _nhwc_flat(x)).view(
                     w.shape[0], w.shape[2], w.shape[3], w.shape[1]
                 ).permute(0, 3, 1, 2)
+                if dw.dtype != ctx.wdtype:
+                    dw = dw.to(ctx.wdtype)
             if ctx.has_bias and ctx.needs_input_grad[2]:
                 db = g2.sum(0, dtype=torch.float32)
-            return dx, dw, db, None, None, None, None
+            base = (dx, dw, db, None, None, None, None)
+            return base + (None, None) if ctx.with_shadows else base
 
         if ctx.needs_input_grad[0]:
             if stride == 1:
-                wt = w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)
+                wt = w if ctx.shadow_bwd else \
+                    w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)
                 dx = ext.conv2d_nhwc_fwd(gout, wt, None, 1, k - 1 - pad, 0, None)
             else:
-                wt = w.transpose(0, 1).contiguous(memory_format=CL)  # (C,K,k,k)
+                wt = w if ctx.shadow_bwd else \
+                    w.transpose(0, 1).contiguous(memory_format=CL)  # (C,K,k,k)
                 dx = ext.conv2d_nhwc_fracstride(
                     gout, wt, None, stride, pad, x.shape[2], x.shape[3], 0, None
                 )
         if ctx.needs_input_grad[1]:
             ws = ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0)
-            dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
-                .contiguous(memory_format=CL)
+            if ctx.wdtype == torch.float32:
+                # (B,R,S,A) contiguous permuted to (K,C,k,k) IS the standard
+                # channels_last layout: the fp32 workspace is the grad, no
+                # bf16 round trip, no copy.
+                dw = ws.permute(0, 3, 1, 2)
+            else:
+                dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
+                    .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _channel_sum(gout)
-        return dx, dw, db, None, None, None, None
+        base = (dx, dw, db, None, None, None, None)
+        return base + (None, None) if ctx.with_shadows else base
 
 
 class ConvT2dNHWCFn(torch.autograd.Function):
@@ -169,10 +250,14 @@ class ConvT2dNHWCFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int = 0,
-                want_stats: bool = False):
+                want_stats: bool = False, w_fwd=None, w_bwd=None):
+        # With shadows: w is the raw fp32 param (receives the grad), w_fwd is
+        # the pre-flipped/transposed (Co,Ci,k,k) bf16 forward form, w_bwd the
+        # plain-cast (Ci,Co,k,k) bf16 form (dgrad + gemm operand).
         ext = _ext()
-        k = w.shape[2]
-        co = w.shape[1]
+        wp = w_bwd if w_bwd is not None else w  # plain (Ci,Co,k,k) bf16 form
+        k = wp.shape[2]
+        co = wp.shape[1]
         n, _, h, wdt = x.shape
         b32 = b.float() if b is not None else None
         stats = (
@@ -184,7 +269,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
             # 1x1 -> kxk: out[n, y, x, co] = sum_ci in[n,ci] w[ci,co,y,x].
             # The channels_last (Ci,Co,k,k) weight is physically (Ci,k,k,Co),
             # so its NHWC flattening is already (Ci, [y,x,co]).
-            out = torch.mm(_nhwc_flat(x), _nhwc_flat(w))
+            out = torch.mm(_nhwc_flat(x), _nhwc_flat(wp))
             if b is not None:
                 out = out.view(n, k * k, co) + b.to(out.dtype)
             out = _act_fwd_torch(out, act)
@@ -192,17 +277,22 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 stats = _gemm_stats(out.reshape(n * k * k, co)).unsqueeze(0)
             out = out.view(n, k, k, co).permute(0, 3, 1, 2)
         elif stride == 1:
-            wt = w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)  # (Co,Ci,k,k)
+            wt = w_fwd if w_fwd is not None else \
+                w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)  # (Co,Ci,k,k)
             out = ext.conv2d_nhwc_fwd(x, wt, b32, 1, k - 1 - pad, act, stats)
         else:
             oh = (h - 1) * stride - 2 * pad + k
             ow = (wdt - 1) * stride - 2 * pad + k
-            wt = w.transpose(0, 1).contiguous(memory_format=CL)  # (Co, Ci, k, k)
+            wt = w_fwd if w_fwd is not None else \
+                w.transpose(0, 1).contiguous(memory_format=CL)  # (Co, Ci, k, k)
             out = ext.conv2d_nhwc_fracstride(x, wt, b32, stride, pad, oh, ow,
                                              act, stats)
-        ctx.save_for_backward(x, w, out if act != 0 else None)
+        ctx.save_for_backward(x, wp, out if act != 0 else None)
+        ctx.shadow_bwd = w_bwd is not None
+        ctx.with_shadows = w_fwd is not None or w_bwd is not None
         ctx.stride, ctx.pad, ctx.has_bias, ctx.gemm = stride, pad, b is not None, gemm
         ctx.act = act
+        ctx.wdtype = w.dtype
         if stats is None:
             stats = torch.empty(0, device=x.device)
         ctx.mark_non_differentiable(stats)
@@ -231,21 +321,28 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 # dwf (Ci, k*k*Co) -> (Ci, k, k, Co) physical = CL (Ci,Co,k,k)
                 dwf = torch.mm(_nhwc_flat(x).t(), g2)
                 dw = dwf.view(x.shape[1], k, k, w.shape[1]).permute(0, 3, 1, 2)
+                if dw.dtype != ctx.wdtype:
+                    dw = dw.to(ctx.wdtype)
             if ctx.has_bias and ctx.needs_input_grad[2]:
                 db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
-            return dx, dw, db, None, None, None, None
+            base = (dx, dw, db, None, None, None, None)
+            return base + (None, None) if ctx.with_shadows else base
 
         if ctx.needs_input_grad[0]:
             # dgrad of convT = plain conv with the untransposed weight
-            wl = w.contiguous(memory_format=CL)  # (Ci, Co, k, k): Co in, Ci out
+            wl = w if ctx.shadow_bwd else w.contiguous(memory_format=CL)
             dx = ext.conv2d_nhwc_fwd(gout, wl, None, stride, pad, 0, None)
         if ctx.needs_input_grad[1]:
             ws = ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad, 0)
-            dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
-                .contiguous(memory_format=CL)
+            if ctx.wdtype == torch.float32:
+                dw = ws.permute(0, 3, 1, 2)
+            else:
+                dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
+                    .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             db = _channel_sum(gout)
-        return dx, dw, db, None, None, None, None
+        base = (dx, dw, db, None, None, None, None)
+        return base + (None, None) if ctx.with_shadows else base
 
 
 def _use_hip_path(x: torch.Tensor) -> bool:
@@ -275,9 +372,10 @@ class Conv2d(nn.Conv2d):
         ):
             with torch.autocast("cuda", enabled=False):
                 xl = _to_cl_bf16(x)
-                wl = _to_cl_bf16(self.weight)
+                sh = _conv_shadows(self)
                 out, _ = Conv2dNHWCFn.apply(
-                    xl, wl, self.bias, self.stride[0], self.padding[0], 0, False
+                    xl, self.weight, self.bias, self.stride[0],
+                    self.padding[0], 0, False, sh["f"], sh["b"]
                 )
                 return out
         return super().forward(x)
@@ -301,9 +399,10 @@ class ConvTranspose2d(nn.ConvTranspose2d):
         ):
             with torch.autocast("cuda", enabled=False):
                 xl = _to_cl_bf16(x)
-                wl = _to_cl_bf16(self.weight)
+                sh = _conv_shadows(self)
                 out, _ = ConvT2dNHWCFn.apply(
-                    xl, wl, self.bias, self.stride[0], self.padding[0], 0, False
+                    xl, self.weight, self.bias, self.stride[0],
+                    self.padding[0], 0, False, sh["f"], sh["b"]
                 )
                 return out
         return super().forward(x, output_size)

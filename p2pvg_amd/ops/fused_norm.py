@@ -69,13 +69,13 @@ class FusedBNActFn(torch.autograd.Function):
 
 def fused_conv_bn_act(x, conv, bn, act: int):
     """conv (with stats epilogue) -> fused BN+act. Training and eval modes."""
-    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _to_cl_bf16
+    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _conv_shadows, _to_cl_bf16
 
     ext = _ext()
     training = bn.training
     with torch.autocast("cuda", enabled=False):
         xl = _to_cl_bf16(x)
-        wl = _to_cl_bf16(conv.weight)
+        sh = _conv_shadows(conv)
         want_stats = training
         # BatchNorm is shift-invariant, so the conv bias has EXACTLY zero
         # effect on the block output; skip it (and its gradient reduction).
@@ -83,11 +83,13 @@ def fused_conv_bn_act(x, conv, bn, act: int):
         # initialized to 0 (init_weights) and stays ~0.
         if isinstance(conv, nn.ConvTranspose2d):
             out, stats = ConvT2dNHWCFn.apply(
-                xl, wl, None, conv.stride[0], conv.padding[0], 0, want_stats
+                xl, conv.weight, None, conv.stride[0], conv.padding[0],
+                0, want_stats, sh["f"], sh["b"]
             )
         else:
             out, stats = Conv2dNHWCFn.apply(
-                xl, wl, None, conv.stride[0], conv.padding[0], 0, want_stats
+                xl, conv.weight, None, conv.stride[0], conv.padding[0],
+                0, want_stats, sh["f"], sh["b"]
             )
         if not want_stats:
             stats = None
@@ -107,18 +109,20 @@ def fused_conv_bn_act(x, conv, bn, act: int):
 
 def fused_conv_act(x, conv, act: int):
     """conv with the activation fused straight into the epilogue (no BN)."""
-    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _to_cl_bf16
+    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _conv_shadows, _to_cl_bf16
 
     with torch.autocast("cuda", enabled=False):
         xl = _to_cl_bf16(x)
-        wl = _to_cl_bf16(conv.weight)
+        sh = _conv_shadows(conv)
         if isinstance(conv, nn.ConvTranspose2d):
             out, _ = ConvT2dNHWCFn.apply(
-                xl, wl, conv.bias, conv.stride[0], conv.padding[0], act, False
+                xl, conv.weight, conv.bias, conv.stride[0], conv.padding[0],
+                act, False, sh["f"], sh["b"]
             )
         else:
             out, _ = Conv2dNHWCFn.apply(
-                xl, wl, conv.bias, conv.stride[0], conv.padding[0], act, False
+                xl, conv.weight, conv.bias, conv.stride[0], conv.padding[0],
+                act, False, sh["f"], sh["b"]
             )
         return out
 
