@@ -130,3 +130,34 @@ def test_multi_cp_and_loop_generation(tiny_cfg):
     loop = gen_cli.loop_generate(model, x[0], x[7], seg_len=4)
     assert len(loop) == 7
     assert torch.equal(loop[0], x[0])
+
+
+def test_conv_weight_shadows_track_weight_version():
+    """The per-step bf16 weight shadows must refill when the weight changes
+    in place (version counter) and hold the right flip/transpose forms."""
+    import torch
+
+    from p2pvg_amd.ops.conv import Conv2d, ConvTranspose2d, _conv_shadows
+
+    torch.manual_seed(0)
+    c = Conv2d(8, 16, 3, stride=1, padding=1, bias=False)
+    sh = _conv_shadows(c)
+    assert sh["f"].dtype == torch.bfloat16
+    torch.testing.assert_close(sh["f"].float(), c.weight.detach().bfloat16().float())
+    torch.testing.assert_close(
+        sh["b"].float(),
+        c.weight.detach().bfloat16().flip(2, 3).transpose(0, 1).float(),
+    )
+
+    with torch.no_grad():
+        c.weight.add_(1.0)
+    sh2 = _conv_shadows(c)
+    assert sh2 is sh, "shadow storage must be reused"
+    torch.testing.assert_close(sh["f"].float(), c.weight.detach().bfloat16().float())
+
+    ct = ConvTranspose2d(8, 4, 4, stride=2, padding=1, bias=False)
+    sht = _conv_shadows(ct)
+    torch.testing.assert_close(
+        sht["f"].float(), ct.weight.detach().bfloat16().transpose(0, 1).float()
+    )
+    torch.testing.assert_close(sht["b"].float(), ct.weight.detach().bfloat16().float())
