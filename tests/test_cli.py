@@ -82,3 +82,25 @@ def test_bench_contract_cpu(tmp_path):
         assert key in d, f"bench JSON missing {key}"
     assert d["data"] == "synthetic"
     assert d["config"]["global_batch"] == 2
+
+
+def test_phase_timing_harness_cpu():
+    """SURVEY §5.1 per-phase timing harness runs on CPU and covers all six
+    phases."""
+    import numpy as np
+    import torch
+
+    from p2pvg_amd.core import Config
+    from p2pvg_amd.models import P2PModel
+    from p2pvg_amd.utils.phases import measure_phase_times
+
+    cfg = Config(dataset="mnist", backbone="dcgan", channels=1, batch_size=2,
+                 max_seq_len=6, g_dim=16, z_dim=4, rnn_size=32, device="cpu")
+    torch.manual_seed(0)
+    np.random.seed(0)
+    model = P2PModel(cfg)
+    x = torch.rand(6, 2, 1, 64, 64)
+    times = measure_phase_times(model, x, iters=1, warmup=1)
+    assert set(times) == {"htod", "forward", "bwd_nonprior", "bwd_prior",
+                          "step_nonprior", "step_prior"}
+    assert all(v >= 0 for v in times.values())
