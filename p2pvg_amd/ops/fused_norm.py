@@ -94,6 +94,11 @@ def fused_conv_bn_act(x, conv, bn, act: int):
         if not want_stats:
             stats = None
         if training:
+            if bn.track_running_stats and bn.num_batches_tracked is not None:
+                # keep nn.BatchNorm2d state semantics (cumulative-average
+                # momentum=None and checkpoint parity); device-side add_ stays
+                # hipGraph-capturable
+                bn.num_batches_tracked.add_(1)
             return FusedBNActFn.apply(
                 out, stats, bn.weight, bn.bias,
                 bn.running_mean if bn.track_running_stats else None,

@@ -79,6 +79,46 @@ class GraphedTrainStep:
         model._backward_and_step(*losses)
         return tuple(v.detach() for v in losses)
 
+    def _optimizers(self):
+        return [v for v in vars(self.model).items()
+                if isinstance(v[1], torch.optim.Optimizer)]
+
+    def _snapshot_train_state(self):
+        """Clone params/buffers + optimizer state so capture warmup applies
+        zero net optimizer steps (warmup repeats the same batch several
+        times; without the restore every new graph key would push up to
+        warmup_iters duplicate-data updates into training)."""
+        model_snap = {k: v.detach().clone()
+                      for k, v in self.model.state_dict().items()}
+        opt_snap = {}
+        for name, opt in self._optimizers():
+            opt_snap[name] = {
+                id(p): {k: (v.detach().clone() if torch.is_tensor(v) else v)
+                        for k, v in st.items()}
+                for p, st in opt.state.items()
+            }
+        return model_snap, opt_snap
+
+    def _restore_train_state(self, snap):
+        """Copy the snapshot back IN PLACE (same storages: Adam state created
+        during warmup stays allocated outside the graph pool; entries that
+        did not exist at snapshot time are reset to the fresh-state zeros)."""
+        model_snap, opt_snap = snap
+        self.model.load_state_dict(model_snap)
+        for name, opt in self._optimizers():
+            saved = opt_snap.get(name, {})
+            for p, st in opt.state.items():
+                sv = saved.get(id(p))
+                for k, v in st.items():
+                    if not torch.is_tensor(v):
+                        if sv is not None and k in sv:
+                            st[k] = sv[k]
+                        continue
+                    if sv is not None and k in sv:
+                        v.copy_(sv[k])
+                    else:
+                        v.zero_()
+
     def _capture(self, plan: StepPlan, prev: torch.Tensor, cur: torch.Tensor,
                  tun: torch.Tensor, dts: torch.Tensor) -> _GraphEntry:
         entry = _GraphEntry()
@@ -88,6 +128,7 @@ class GraphedTrainStep:
         entry.dts = dts.clone()
         entry.loss_out = torch.zeros(4, device=prev.device)
 
+        snap = self._snapshot_train_state()
         # warmup on a side stream (standard CUDAGraph recipe)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -96,6 +137,8 @@ class GraphedTrainStep:
                 self._inner(entry, plan)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
+        self._restore_train_state(snap)
+        del snap
         # release warmup's cached segments: expandable_segments is a no-op on
         # this ROCm build, and the graph's private pool cannot reuse the
         # allocator's fragmented free blocks (observed: 116 GB reserved but
