@@ -22,8 +22,67 @@ from p2pvg_amd.utils.image import make_grid, save_gif, save_image, to_uint8_hwc
 from p2pvg_amd.utils.vis import add_gt_cp_border, add_samples_cp_border
 
 
+def _read_video_mp4(vid_name: str) -> list:
+    """Decode an .mp4/.avi/.mov (reference generate.py:29-39 reads .mp4 via
+    imageio/ffmpeg). Tries, in order: imageio(.v3), PyAV, an ffmpeg binary
+    (rawvideo pipe). All are optional dependencies; a clear error names them
+    when none is present."""
+    frames = []
+    try:  # imageio v3 or v2 (bundles imageio-ffmpeg when installed)
+        try:
+            import imageio.v3 as iio
+
+            for fr in iio.imiter(vid_name, plugin="pyav"):
+                frames.append(np.asarray(fr, dtype=np.float32) / 255.0)
+        except Exception:
+            import imageio
+
+            for fr in imageio.get_reader(vid_name):
+                frames.append(np.asarray(fr, dtype=np.float32) / 255.0)
+        if frames:
+            return frames
+    except ImportError:
+        pass
+    try:  # PyAV directly
+        import av
+
+        with av.open(vid_name) as container:
+            for fr in container.decode(video=0):
+                frames.append(
+                    np.asarray(fr.to_rgb().to_ndarray(), dtype=np.float32) / 255.0
+                )
+        if frames:
+            return frames
+    except ImportError:
+        pass
+    import shutil
+    import subprocess
+
+    ffmpeg = shutil.which("ffmpeg")
+    ffprobe = shutil.which("ffprobe")
+    if ffmpeg and ffprobe:
+        probe = subprocess.run(
+            [ffprobe, "-v", "error", "-select_streams", "v:0", "-show_entries",
+             "stream=width,height", "-of", "csv=p=0", vid_name],
+            capture_output=True, text=True, check=True,
+        )
+        w, h = (int(v) for v in probe.stdout.strip().split(",")[:2])
+        raw = subprocess.run(
+            [ffmpeg, "-v", "error", "-i", vid_name, "-f", "rawvideo",
+             "-pix_fmt", "rgb24", "-"],
+            capture_output=True, check=True,
+        ).stdout
+        arr = np.frombuffer(raw, dtype=np.uint8).reshape(-1, h, w, 3)
+        return [f.astype(np.float32) / 255.0 for f in arr]
+    raise RuntimeError(
+        f"cannot decode {vid_name!r}: no mp4 backend available. Install one of "
+        "imageio[ffmpeg], av (PyAV), or an ffmpeg binary on PATH — or pass a "
+        ".gif / frame directory instead."
+    )
+
+
 def read_video(vid_name: str) -> torch.Tensor:
-    """Read a video file or frame directory -> (t, 1, c, h, w) in [0,1]."""
+    """Read a video (.mp4/.gif), or a frame directory -> (t, 1, c, h, w) in [0,1]."""
     frames = []
     if os.path.isdir(vid_name):
         from PIL import Image
@@ -32,6 +91,8 @@ def read_video(vid_name: str) -> torch.Tensor:
             if f.lower().endswith((".png", ".jpg", ".jpeg")):
                 with Image.open(os.path.join(vid_name, f)) as im:
                     frames.append(np.asarray(im.convert("RGB"), dtype=np.float32) / 255.0)
+    elif vid_name.lower().endswith((".mp4", ".avi", ".mov", ".mkv", ".webm")):
+        frames = _read_video_mp4(vid_name)
     else:
         from PIL import Image, ImageSequence
 
