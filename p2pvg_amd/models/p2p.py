@@ -302,6 +302,27 @@ class P2PModel(nn.Module):
         prior = list(self.prior.parameters())
         return nonprior, prior
 
+    def zero_grads(self):
+        """Zero every param grad in ONE fused dispatch group.
+
+        Replaces nn.Module.zero_grad(set_to_none=False)'s per-tensor fills
+        (~200 tiny kernels per step). Grads are materialized once with the
+        params' own memory format and then kept at stable addresses — the
+        custom conv/BN backward kernels ACCUMULATE into these buffers in
+        kernel (ops/conv.py::_acc_target), so autograd's per-use accumulation
+        adds never dispatch, and hipGraph capture sees fixed pointers."""
+        gs = getattr(self, "_grad_list", None)
+        if gs is None:
+            for p in self.parameters():
+                if p.grad is None:
+                    p.grad = torch.zeros_like(p)
+            self._grad_list = gs = [p.grad for p in self.parameters()]
+        torch._foreach_zero_(gs)
+
+    def zero_grad(self, set_to_none: bool = True):
+        # keep the stable-buffer discipline whatever the caller passes
+        self.zero_grads()
+
     def _backward_and_step(self, mse_loss, kld_loss, cpc_loss, align_loss):
         cfg = self.cfg
         mode = getattr(cfg, "backward_mode", "pruned")
@@ -320,26 +341,38 @@ class P2PModel(nn.Module):
         # pre-step graph, then step. The effective update rule is unchanged:
         # non-prior params move by -lr*Adam(dL1), prior by -lr*Adam(dL2).
 
+        from ..ops.conv import no_weight_grads
+
         if mode == "reference":
             # two full-graph traversals, exactly as the reference pays them
             loss1.backward(retain_graph=True)
             stash = [p.grad.clone() if p.grad is not None else None for p in nonprior]
-            self.prior.zero_grad()
+            for p in self.prior.parameters():
+                if p.grad is not None:
+                    p.grad.zero_()
             loss2.backward()
             # the reference stepped non-prior before loss2.backward, so the
             # L2 grads that leak into non-prior params never affect updates;
-            # restore the phase-1 grads to reproduce that.
+            # restore the phase-1 grads to reproduce that. copy_ (not rebind)
+            # keeps grad buffer addresses stable for the fused zero/Adam.
             for p, g in zip(nonprior, stash):
                 if g is not None:
-                    p.grad = g
+                    if p.grad is None:
+                        p.grad = g
+                    else:
+                        p.grad.copy_(g)
         else:
             # phase 1: grads of loss1 into everything but the prior
             torch.autograd.backward(loss1, inputs=nonprior, retain_graph=True)
             # phase 2: grads of loss2 into the prior only — autograd prunes
             # the traversal to the kld->prior and cpc->prior paths (one
             # decoder dgrad chain at the cp step + the prior BPTT chain)
-            # instead of re-walking the whole unrolled graph.
-            torch.autograd.backward(loss2, inputs=prior)
+            # instead of re-walking the whole unrolled graph. The pruned
+            # traversal still reaches the decoder/predictor backward nodes
+            # (their dx is on the path); no_weight_grads suppresses their
+            # discarded-anyway weight-gradient work.
+            with no_weight_grads():
+                torch.autograd.backward(loss2, inputs=prior)
 
         if grad_sync is not None:
             grad_sync.sync_nonprior()

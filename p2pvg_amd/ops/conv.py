@@ -26,6 +26,47 @@ import torch.nn as nn
 
 CL = torch.channels_last
 
+# Phase flag for the two-phase backward: during phase 2 (prior-only update)
+# autograd still traverses the decoder/predictor chain on the cpc path and
+# reports needs_input_grad[weight]=True, but those weight grads are discarded
+# by the update rule (reference zeroes them at the next step start). Skipping
+# them here saves the wasted wgrad kernels AND keeps the managed .grad
+# buffers exactly equal to the phase-1 gradients Adam consumes.
+_WEIGHT_GRADS = True
+
+
+class no_weight_grads:
+    """Context manager: suppress weight/bias/BN-param gradient work in the
+    custom backward Functions (used around the phase-2 backward)."""
+
+    def __enter__(self):
+        global _WEIGHT_GRADS
+        self._prev = _WEIGHT_GRADS
+        _WEIGHT_GRADS = False
+        return self
+
+    def __exit__(self, *a):
+        global _WEIGHT_GRADS
+        _WEIGHT_GRADS = self._prev
+        return False
+
+
+def weight_grads_enabled() -> bool:
+    return _WEIGHT_GRADS
+
+
+def _acc_target(p):
+    """The managed fp32 .grad buffer of a param, if accumulate-in-kernel is
+    possible (buffer exists — materialized by P2PModel.zero_grads — dense
+    fp32, matching element count). None -> caller falls back to returning
+    the grad through autograd's accumulation."""
+    if p is None or p.dtype != torch.float32:
+        return None
+    g = p.grad
+    if g is None or g.dtype != torch.float32 or g.numel() != p.numel():
+        return None
+    return g
+
 
 def _ext():
     from . import _hip_ext_loader
@@ -72,15 +113,23 @@ def _gemm_stats(out2):
     return torch.stack([f.sum(0), (f * f).sum(0)]).contiguous()
 
 
-def _channel_sum(gout):
+def _channel_sum(gout, acc=None):
     """db = sum over (N,H,W). ATen's strided (0,2,3) reduction on channels_last
-    runs ~8x off bandwidth; use the NHWC channel-sum kernel when eligible."""
+    runs ~8x off bandwidth; use the NHWC channel-sum kernel when eligible.
+    With acc (the bias .grad): accumulate in kernel, return None."""
     if (gout.is_cuda and gout.dtype == torch.bfloat16
             and gout.is_contiguous(memory_format=CL)):
         C = gout.shape[1]
         if C <= 4 or (C % 8 == 0 and 256 % (C // 8) == 0):
+            if acc is not None:
+                _ext().channel_sum_nhwc(gout, acc)
+                return None
             return _ext().channel_sum_nhwc(gout)
-    return gout.sum(dim=(0, 2, 3), dtype=torch.float32)
+    db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
+    if acc is not None:
+        acc.add_(db)
+        return None
+    return db
 
 
 @torch.no_grad()
@@ -156,10 +205,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
         wf = w_fwd if w_fwd is not None else w
         k = wf.shape[2]
         b32 = b.float() if b is not None else None
-        stats = (
-            torch.zeros(64, 2, wf.shape[0], device=x.device, dtype=torch.float32)
-            if want_stats else None
-        )
+        stats = None
         # degenerate whole-image conv (k == H, pad 0): plain GEMM
         gemm = pad == 0 and k == x.shape[2] and k == x.shape[3]
         if gemm:
@@ -175,7 +221,8 @@ class Conv2dNHWCFn(torch.autograd.Function):
             ctx.save_for_backward(x, wf, out if act != 0 else None)
             ctx.shadow_bwd = False
         else:
-            out = ext.conv2d_nhwc_fwd(x, wf, b32, stride, pad, act, stats)
+            out, stats = ext.conv2d_nhwc_fwd(x, wf, b32, stride, pad, act,
+                                             want_stats)
             wb = w_bwd if w_bwd is not None else wf
             ctx.save_for_backward(x, wb, out if act != 0 else None)
             ctx.shadow_bwd = w_bwd is not None
@@ -183,7 +230,8 @@ class Conv2dNHWCFn(torch.autograd.Function):
         ctx.stride, ctx.pad, ctx.has_bias, ctx.gemm = stride, pad, b is not None, gemm
         ctx.act = act
         ctx.wdtype = w.dtype
-        if stats is None:
+        ctx.wref, ctx.bref = w, b
+        if stats is None or not want_stats:
             stats = torch.empty(0, device=x.device)
         ctx.mark_non_differentiable(stats)
         return out, stats
@@ -200,6 +248,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
         if ctx.act != 0:
             gout = _act_bwd_from_y(gout, y, ctx.act).contiguous(memory_format=CL)
 
+        wge = weight_grads_enabled()
         dx = dw = db = None
         if ctx.gemm:
             g2 = gout.reshape(gout.shape[0], gout.shape[1])  # (N, K)
@@ -207,13 +256,13 @@ class Conv2dNHWCFn(torch.autograd.Function):
                 dx = torch.mm(g2, _nhwc_flat(w)).view(
                     x.shape[0], x.shape[2], x.shape[3], x.shape[1]
                 ).permute(0, 3, 1, 2)
-            if ctx.needs_input_grad[1]:
+            if ctx.needs_input_grad[1] and wge:
                 dw = torch.mm(g2.t(), _nhwc_flat(x)).view(
                     w.shape[0], w.shape[2], w.shape[3], w.shape[1]
                 ).permute(0, 3, 1, 2)
                 if dw.dtype != ctx.wdtype:
                     dw = dw.to(ctx.wdtype)
-            if ctx.has_bias and ctx.needs_input_grad[2]:
+            if ctx.has_bias and ctx.needs_input_grad[2] and wge:
                 db = g2.sum(0, dtype=torch.float32)
             base = (dx, dw, db, None, None, None, None)
             return base + (None, None) if ctx.with_shadows else base
@@ -222,25 +271,34 @@ class Conv2dNHWCFn(torch.autograd.Function):
             if stride == 1:
                 wt = w if ctx.shadow_bwd else \
                     w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)
-                dx = ext.conv2d_nhwc_fwd(gout, wt, None, 1, k - 1 - pad, 0, None)
+                dx = ext.conv2d_nhwc_fwd(gout, wt, None, 1, k - 1 - pad, 0,
+                                         False)[0]
             else:
                 wt = w if ctx.shadow_bwd else \
                     w.transpose(0, 1).contiguous(memory_format=CL)  # (C,K,k,k)
                 dx = ext.conv2d_nhwc_fracstride(
-                    gout, wt, None, stride, pad, x.shape[2], x.shape[3], 0, None
-                )
-        if ctx.needs_input_grad[1]:
-            ws = ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0)
-            if ctx.wdtype == torch.float32:
-                # (B,R,S,A) contiguous permuted to (K,C,k,k) IS the standard
-                # channels_last layout: the fp32 workspace is the grad, no
-                # bf16 round trip, no copy.
-                dw = ws.permute(0, 3, 1, 2)
+                    gout, wt, None, stride, pad, x.shape[2], x.shape[3], 0,
+                    False
+                )[0]
+        if ctx.needs_input_grad[1] and wge:
+            wg = _acc_target(ctx.wref)
+            if wg is not None and ctx.wdtype == torch.float32 \
+                    and wg.is_contiguous(memory_format=CL):
+                # accumulate straight into the managed fp32 .grad: autograd's
+                # per-use AccumulateGrad adds never run for conv weights
+                ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0, wg)
             else:
-                dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
-                    .contiguous(memory_format=CL)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = _channel_sum(gout)
+                ws = ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0)
+                if ctx.wdtype == torch.float32:
+                    # (B,R,S,A) contiguous permuted to (K,C,k,k) IS the
+                    # standard channels_last layout: the fp32 workspace is
+                    # the grad, no bf16 round trip, no copy.
+                    dw = ws.permute(0, 3, 1, 2)
+                else:
+                    dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
+                        .contiguous(memory_format=CL)
+        if ctx.has_bias and ctx.needs_input_grad[2] and wge:
+            db = _channel_sum(gout, _acc_target(ctx.bref))
         base = (dx, dw, db, None, None, None, None)
         return base + (None, None) if ctx.with_shadows else base
 
@@ -260,10 +318,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         co = wp.shape[1]
         n, _, h, wdt = x.shape
         b32 = b.float() if b is not None else None
-        stats = (
-            torch.zeros(64, 2, co, device=x.device, dtype=torch.float32)
-            if want_stats else None
-        )
+        stats = None
         gemm = stride == 1 and pad == 0 and h == 1 and wdt == 1
         if gemm:
             # 1x1 -> kxk: out[n, y, x, co] = sum_ci in[n,ci] w[ci,co,y,x].
@@ -279,21 +334,23 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         elif stride == 1:
             wt = w_fwd if w_fwd is not None else \
                 w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)  # (Co,Ci,k,k)
-            out = ext.conv2d_nhwc_fwd(x, wt, b32, 1, k - 1 - pad, act, stats)
+            out, stats = ext.conv2d_nhwc_fwd(x, wt, b32, 1, k - 1 - pad, act,
+                                             want_stats)
         else:
             oh = (h - 1) * stride - 2 * pad + k
             ow = (wdt - 1) * stride - 2 * pad + k
             wt = w_fwd if w_fwd is not None else \
                 w.transpose(0, 1).contiguous(memory_format=CL)  # (Co, Ci, k, k)
-            out = ext.conv2d_nhwc_fracstride(x, wt, b32, stride, pad, oh, ow,
-                                             act, stats)
+            out, stats = ext.conv2d_nhwc_fracstride(x, wt, b32, stride, pad,
+                                                    oh, ow, act, want_stats)
         ctx.save_for_backward(x, wp, out if act != 0 else None)
         ctx.shadow_bwd = w_bwd is not None
         ctx.with_shadows = w_fwd is not None or w_bwd is not None
         ctx.stride, ctx.pad, ctx.has_bias, ctx.gemm = stride, pad, b is not None, gemm
         ctx.act = act
         ctx.wdtype = w.dtype
-        if stats is None:
+        ctx.wref, ctx.bref = w, b
+        if stats is None or not want_stats:
             stats = torch.empty(0, device=x.device)
         ctx.mark_non_differentiable(stats)
         return out, stats
@@ -310,6 +367,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         if ctx.act != 0:
             gout = _act_bwd_from_y(gout, y, ctx.act).contiguous(memory_format=CL)
 
+        wge = weight_grads_enabled()
         dx = dw = db = None
         if ctx.gemm:
             n = x.shape[0]
@@ -317,13 +375,13 @@ class ConvT2dNHWCFn(torch.autograd.Function):
             wf = _nhwc_flat(w)     # (Ci, k*k*Co)
             if ctx.needs_input_grad[0]:
                 dx = torch.mm(g2, wf.t()).view(n, 1, 1, x.shape[1]).permute(0, 3, 1, 2)
-            if ctx.needs_input_grad[1]:
+            if ctx.needs_input_grad[1] and wge:
                 # dwf (Ci, k*k*Co) -> (Ci, k, k, Co) physical = CL (Ci,Co,k,k)
                 dwf = torch.mm(_nhwc_flat(x).t(), g2)
                 dw = dwf.view(x.shape[1], k, k, w.shape[1]).permute(0, 3, 1, 2)
                 if dw.dtype != ctx.wdtype:
                     dw = dw.to(ctx.wdtype)
-            if ctx.has_bias and ctx.needs_input_grad[2]:
+            if ctx.has_bias and ctx.needs_input_grad[2] and wge:
                 db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
             base = (dx, dw, db, None, None, None, None)
             return base + (None, None) if ctx.with_shadows else base
@@ -331,16 +389,21 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # dgrad of convT = plain conv with the untransposed weight
             wl = w if ctx.shadow_bwd else w.contiguous(memory_format=CL)
-            dx = ext.conv2d_nhwc_fwd(gout, wl, None, stride, pad, 0, None)
-        if ctx.needs_input_grad[1]:
-            ws = ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad, 0)
-            if ctx.wdtype == torch.float32:
-                dw = ws.permute(0, 3, 1, 2)
+            dx = ext.conv2d_nhwc_fwd(gout, wl, None, stride, pad, 0, False)[0]
+        if ctx.needs_input_grad[1] and wge:
+            wg = _acc_target(ctx.wref)
+            if wg is not None and ctx.wdtype == torch.float32 \
+                    and wg.is_contiguous(memory_format=CL):
+                ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad, 0, wg)
             else:
-                dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
-                    .contiguous(memory_format=CL)
-        if ctx.has_bias and ctx.needs_input_grad[2]:
-            db = _channel_sum(gout)
+                ws = ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad, 0)
+                if ctx.wdtype == torch.float32:
+                    dw = ws.permute(0, 3, 1, 2)
+                else:
+                    dw = ws.permute(0, 3, 1, 2).to(torch.bfloat16) \
+                        .contiguous(memory_format=CL)
+        if ctx.has_bias and ctx.needs_input_grad[2] and wge:
+            db = _channel_sum(gout, _acc_target(ctx.bref))
         base = (dx, dw, db, None, None, None, None)
         return base + (None, None) if ctx.with_shadows else base
 

@@ -338,16 +338,21 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
     // all 64 lanes reach the shuffles (no divergent early-out above)
     if (a.stats != nullptr) {
       // reduce the 4 lanes sharing this column (l, l+16, l+32, l+48), then
-      // one atomic per column into a per-block bucket (64 copies cut the
-      // same-address contention that serializes large-M launches)
+      // STORE the per-(block, wave-pair) partial into its own bucket row:
+      // no atomics, no pre-zeroing, and the serial bucket walk in
+      // bn_finalize makes the statistics bitwise-deterministic.
       csum += __shfl_xor(csum, 16, 64);
       csum += __shfl_xor(csum, 32, 64);
       csq += __shfl_xor(csq, 16, 64);
       csq += __shfl_xor(csq, 32, 64);
       if ((lane >> 4) == 0 && colv) {
-        float* bucket = a.stats + (long)(blockIdx.x & 63) * 2 * a.K;
-        atomicAdd(&bucket[col], csum);
-        atomicAdd(&bucket[a.K + col], csq);
+        // waves (0,1) and (2,3) cover disjoint col halves x disjoint M
+        // halves; 2 rows per block make every (row, col) written exactly once
+        const long row =
+            ((long)blockIdx.z * a.mblocks + blockIdx.x) * 2 + (wid >> 1);
+        float* slot = a.stats + row * 2 * a.K;
+        slot[col] = csum;
+        slot[a.K + col] = csq;
       }
     }
   }
@@ -356,9 +361,17 @@ __global__ __launch_bounds__(THREADS) void conv2d_nhwc_fwd_kernel(
 template <int BM, int BN, int BK, int KSIZE, int STRIDE, bool FRAC, bool RSCLIN>
 void launch_one(const torch::Tensor& in, const torch::Tensor& w,
                 const c10::optional<torch::Tensor>& bias, torch::Tensor& out,
-                ConvArgs& a, int nz) {
+                ConvArgs& a, int nz, bool want_stats,
+                torch::Tensor& stats_out) {
   const int M = a.Nb * a.HO * a.WO;
   a.mblocks = ceil_div(M, BM);
+  if (want_stats) {
+    // one bucket row per (block, wave-pair): written by STORES in the
+    // epilogue — never zeroed, combined serially in bn_finalize
+    stats_out = torch::empty({(long)a.mblocks * nz * 2, 2, a.K},
+                             in.options().dtype(torch::kFloat32));
+    a.stats = stats_out.data_ptr<float>();
+  }
   dim3 grid(a.mblocks, ceil_div(a.K, BN), nz);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(
@@ -374,7 +387,8 @@ void launch_one(const torch::Tensor& in, const torch::Tensor& w,
 template <int KSIZE, int STRIDE, bool FRAC>
 void dispatch_tile(const torch::Tensor& in, const torch::Tensor& w,
                    const c10::optional<torch::Tensor>& bias,
-                   torch::Tensor& out, ConvArgs& a, int nz) {
+                   torch::Tensor& out, ConvArgs& a, int nz, bool want_stats,
+                   torch::Tensor& stats_out) {
   const bool rsclin = !FRAC && a.C < 32 && KSIZE > 1;
   const int M = a.Nb * a.HO * a.WO;
   const bool small = (long)ceil_div(M, 128) * ceil_div(a.K, 128) < 160;
@@ -385,20 +399,20 @@ void dispatch_tile(const torch::Tensor& in, const torch::Tensor& w,
   const bool deep = false;
   if (rsclin) {
     if (narrow)
-      launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+      launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz, want_stats, stats_out);
     else
-      launch_one<128, 128, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz);
+      launch_one<128, 128, 64, KSIZE, STRIDE, FRAC, true>(in, w, bias, out, a, nz, want_stats, stats_out);
   } else if (small) {
-    launch_one<64, 64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    launch_one<64, 64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz, want_stats, stats_out);
   } else if (narrow) {
     if (deep)
-      launch_one<128, 64, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+      launch_one<128, 64, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz, want_stats, stats_out);
     else
-      launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+      launch_one<128, 64, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz, want_stats, stats_out);
   } else if (deep) {
-    launch_one<128, 128, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    launch_one<128, 128, 128, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz, want_stats, stats_out);
   } else {
-    launch_one<128, 128, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz);
+    launch_one<128, 128, 64, KSIZE, STRIDE, FRAC, false>(in, w, bias, out, a, nz, want_stats, stats_out);
   }
 }
 
@@ -415,10 +429,10 @@ void check_nhwc_bf16(const torch::Tensor& t, const char* name) {
 // bias: (K) fp32 optional. act: 0 none, 1 leaky(0.2), 2 tanh, 3 sigmoid.
 // stats: optional (2,K) fp32 ZEROED buffer accumulating sum/sumsq of the
 // activated output per channel (for fused BatchNorm statistics).
-torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
-                              c10::optional<torch::Tensor> bias, long stride,
-                              long pad, long act,
-                              c10::optional<torch::Tensor> stats) {
+std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
+                                            c10::optional<torch::Tensor> bias,
+                                            long stride, long pad, long act,
+                                            bool want_stats) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
@@ -439,22 +453,23 @@ torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
   a.HO = HO; a.WO = WO; a.OH = HO; a.OW = WO;
   a.act = (int)act; a.wk = R; a.oys = 1;
   a.padh[0] = (int)pad; a.padw[0] = (int)pad;
-  a.stats = stats.has_value() ? stats->data_ptr<float>() : nullptr;
+  a.stats = nullptr;
+  torch::Tensor stats_out;
 
   if (R == 3 && stride == 1) {
-    dispatch_tile<3, 1, false>(in, w, bias, out, a, 1);
+    dispatch_tile<3, 1, false>(in, w, bias, out, a, 1, want_stats, stats_out);
   } else if (R == 4 && stride == 2) {
-    dispatch_tile<4, 2, false>(in, w, bias, out, a, 1);
+    dispatch_tile<4, 2, false>(in, w, bias, out, a, 1, want_stats, stats_out);
   } else if (R == 4 && stride == 1) {
-    dispatch_tile<4, 1, false>(in, w, bias, out, a, 1);
+    dispatch_tile<4, 1, false>(in, w, bias, out, a, 1, want_stats, stats_out);
   } else if (R == 2 && stride == 1) {
-    dispatch_tile<2, 1, false>(in, w, bias, out, a, 1);
+    dispatch_tile<2, 1, false>(in, w, bias, out, a, 1, want_stats, stats_out);
   } else if (R == 1 && stride == 1) {
-    dispatch_tile<1, 1, false>(in, w, bias, out, a, 1);
+    dispatch_tile<1, 1, false>(in, w, bias, out, a, 1, want_stats, stats_out);
   } else {
     TORCH_CHECK(false, "unsupported conv geometry: k=", R, " stride=", stride);
   }
-  return out;
+  return {out, want_stats ? stats_out : torch::Tensor()};
 }
 
 // Fractionally-strided convolution (ConvTranspose fwd / stride-s dgrad):
@@ -462,11 +477,10 @@ torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
 // grid.z = s^2 parities. `w` is the FULL (K, C, wk, wk) channels_last weight
 // (already arranged so dim0 = output channels); taps are remapped in-kernel.
 // up_pad is the fractional-conv padding (the original conv's pad).
-torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
-                                     c10::optional<torch::Tensor> bias,
-                                     long up_stride, long up_pad, long OH,
-                                     long OW, long act,
-                                     c10::optional<torch::Tensor> stats) {
+std::vector<torch::Tensor> conv2d_nhwc_fracstride(
+    torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    long up_stride, long up_pad, long OH, long OW, long act,
+    bool want_stats) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
@@ -486,7 +500,8 @@ torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
   a.Nb = Nb; a.H = H; a.W = W; a.C = C; a.K = K;
   a.OH = (int)OH; a.OW = (int)OW;
   a.act = (int)act; a.wk = WK; a.oys = st;
-  a.stats = stats.has_value() ? stats->data_ptr<float>() : nullptr;
+  a.stats = nullptr;
+  torch::Tensor stats_out;
 
   // taps per parity: r with (p + pad - r) % st == 0, offset (p+pad-r)/st
   int ktaps = -1;
@@ -535,6 +550,6 @@ torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
   a.HO = HOc; a.WO = WOc;
 
   TORCH_CHECK(ktaps == 2, "expected 2 taps per axis for k4s2");
-  dispatch_tile<2, 1, true>(in, w, bias, out, a, st * st);
-  return out;
+  dispatch_tile<2, 1, true>(in, w, bias, out, a, st * st, want_stats, stats_out);
+  return {out, want_stats ? stats_out : torch::Tensor()};
 }

@@ -209,7 +209,12 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
     }
   }
 
-  // epilogue: atomic accumulate into the fp32 workspace
+  // epilogue: STORE this slab's partial into its own workspace plane —
+  // every (b,r,s,a) element is written by exactly one block per slab, so
+  // there are no atomics and no pre-zeroing, and the serial slab walk in
+  // wgrad_combine makes the weight gradient bitwise-deterministic at any
+  // split factor.
+  float* slab = ws + (long)blockIdx.z * B * R * S * A;
 #pragma unroll
   for (int j = 0; j < FRA; ++j) {
     const int a = a0 + wn + j * 16 + (lane & 15);
@@ -220,19 +225,36 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
       for (int v = 0; v < 4; ++v) {
         const int b = b0 + wm + i * 16 + (lane >> 4) * 4 + v;
         if (b < B) {
-          atomicAdd(&ws[(((long)b * R + r) * S + s) * A + a], acc[i][j][v]);
+          slab[(((long)b * R + r) * S + s) * A + a] = acc[i][j][v];
         }
       }
     }
   }
 }
 
+// out[e] (+)= sum over slabs ws[z][e] — serial over z (deterministic).
+__global__ __launch_bounds__(256) void wgrad_combine_kernel(
+    const float* __restrict__ ws, float* __restrict__ out, int sp, long E,
+    int accumulate) {
+  for (long e = (long)blockIdx.x * 256 + threadIdx.x; e < E;
+       e += (long)gridDim.x * 256) {
+    float t = 0.f;
+    for (int z = 0; z < sp; ++z) t += ws[(long)z * E + e];
+    out[e] = accumulate ? out[e] + t : t;
+  }
+}
+
 }  // namespace
 
 // Y: channels_last (N,B,HO,WO) bf16; X: channels_last (N,A,H,W) bf16.
-// Returns dW workspace (B, R, S, A) fp32. splitp<=0 -> auto.
+// Returns dW (B, R, S, A) fp32. splitp<=0 -> auto. When `acc` is given it
+// must be an fp32 tensor whose dense layout is (B,R,S,A) (e.g. the conv
+// weight's channels_last .grad): the combine ACCUMULATES into it in place
+// and returns it — this is how the framework skips autograd's per-use grad
+// accumulation adds entirely (the ~29 uses per weight per step land here).
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
-                                long S, long stride, long pad, long splitp) {
+                                long S, long stride, long pad, long splitp,
+                                c10::optional<torch::Tensor> acc) {
   TORCH_CHECK(Y.is_cuda() && Y.scalar_type() == torch::kBFloat16 &&
                   Y.is_contiguous(at::MemoryFormat::ChannelsLast),
               "Y must be bf16 channels_last GPU");
@@ -242,9 +264,7 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
   const int Nb = Y.size(0), B = Y.size(1), HO = Y.size(2), WO = Y.size(3);
   const int A = X.size(1), H = X.size(2), W = X.size(3);
   TORCH_CHECK(X.size(0) == Nb, "batch mismatch");
-
-  auto ws = torch::zeros({B, (long)R, (long)S, A},
-                         Y.options().dtype(torch::kFloat32));
+  const long E = (long)B * R * S * A;
 
   const int BTB = B >= 128 ? 128 : 64;
   const int BTA = A >= 128 ? 128 : 64;
@@ -259,6 +279,10 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
       ceil_div(ceil_div(p_total, sp), PCH) * PCH;  // chunk-aligned
   sp = ceil_div(p_total, p_per_slab);
 
+  // per-slab planes, written by stores — never zeroed
+  auto ws = torch::empty({sp, B, (long)R, (long)S, A},
+                         Y.options().dtype(torch::kFloat32));
+
   dim3 grid(bt * at, (int)(R * S), sp);
   auto stream = at::cuda::getCurrentCUDAStream();
 #define WGRAD_LAUNCH(BB, AA)                                                   \
@@ -272,5 +296,23 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
   else if (BTA == 128) WGRAD_LAUNCH(64, 128);
   else WGRAD_LAUNCH(64, 64);
 #undef WGRAD_LAUNCH
-  return ws;
+
+  torch::Tensor out;
+  int accumulate = 0;
+  if (acc.has_value()) {
+    out = acc.value();
+    TORCH_CHECK(out.scalar_type() == torch::kFloat32 &&
+                    out.is_non_overlapping_and_dense() && out.numel() == E,
+                "wgrad acc: need dense fp32 tensor with B*R*S*A elements");
+    accumulate = 1;
+  } else {
+    if (sp == 1) return ws[0];
+    out = torch::empty({B, (long)R, (long)S, A},
+                       Y.options().dtype(torch::kFloat32));
+  }
+  const int cgrid = (int)std::min<long>(2048, (E + 255) / 256);
+  hipLaunchKernelGGL(wgrad_combine_kernel, dim3(cgrid), dim3(256), 0, stream,
+                     ws.data_ptr<float>(), out.data_ptr<float>(), sp, E,
+                     accumulate);
+  return out;
 }

@@ -25,15 +25,13 @@ std::vector<torch::Tensor> gaussian_kl_bwd(torch::Tensor mu1, torch::Tensor lv1,
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor tr16_probe(long mode);
-torch::Tensor conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
-                              c10::optional<torch::Tensor> bias, long stride,
-                              long pad, long act,
-                              c10::optional<torch::Tensor> stats);
-torch::Tensor conv2d_nhwc_fracstride(torch::Tensor in, torch::Tensor w,
-                                     c10::optional<torch::Tensor> bias,
-                                     long up_stride, long up_pad, long OH,
-                                     long OW, long act,
-                                     c10::optional<torch::Tensor> stats);
+std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
+                                            c10::optional<torch::Tensor> bias,
+                                            long stride, long pad, long act,
+                                            bool want_stats);
+std::vector<torch::Tensor> conv2d_nhwc_fracstride(
+    torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    long up_stride, long up_pad, long OH, long OW, long act, bool want_stats);
 std::vector<torch::Tensor> bn_act_fwd_train(
     torch::Tensor x, torch::Tensor stats, torch::Tensor gamma,
     torch::Tensor beta, c10::optional<torch::Tensor> running_mean,
@@ -45,38 +43,52 @@ torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
 std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                                       torch::Tensor mean, torch::Tensor invstd,
                                       torch::Tensor gamma, torch::Tensor beta,
-                                      torch::Tensor scale, long act);
+                                      torch::Tensor scale, long act,
+                                      c10::optional<torch::Tensor> dgamma_acc,
+                                      c10::optional<torch::Tensor> dbeta_acc);
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
-                                long S, long stride, long pad, long splitp);
+                                long S, long stride, long pad, long splitp,
+                                c10::optional<torch::Tensor> acc);
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor in);
 torch::Tensor maxpool2x2_bwd(torch::Tensor gout, torch::Tensor idx, long H,
                              long W);
 torch::Tensor upsample2x_fwd(torch::Tensor in);
 torch::Tensor upsample2x_bwd(torch::Tensor gout);
-torch::Tensor channel_sum_nhwc(torch::Tensor x);
+torch::Tensor channel_sum_nhwc(torch::Tensor x,
+                               c10::optional<torch::Tensor> acc);
 torch::Tensor sqdiff_sum(torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd,
-        "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA)",
+        "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA); returns (out, stats)",
         pybind11::arg("in"), pybind11::arg("w"), pybind11::arg("bias"),
         pybind11::arg("stride"), pybind11::arg("pad"), pybind11::arg("act"),
-        pybind11::arg("stats") = c10::nullopt);
+        pybind11::arg("want_stats") = false);
   m.def("conv2d_nhwc_fracstride", &conv2d_nhwc_fracstride,
         "fractionally-strided conv, in-kernel parity loop (ConvT fwd / s2 dgrad)",
         pybind11::arg("in"), pybind11::arg("w"), pybind11::arg("bias"),
         pybind11::arg("up_stride"), pybind11::arg("up_pad"), pybind11::arg("OH"),
         pybind11::arg("OW"), pybind11::arg("act"),
-        pybind11::arg("stats") = c10::nullopt);
+        pybind11::arg("want_stats") = false);
   m.def("bn_act_fwd_train", &bn_act_fwd_train, "fused BN+act train fwd (gfx950)");
   m.def("bn_act_fwd_eval", &bn_act_fwd_eval, "fused BN+act eval fwd (gfx950)");
-  m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act bwd (gfx950)");
+  m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act bwd (gfx950, deterministic)",
+        pybind11::arg("x"), pybind11::arg("dy"), pybind11::arg("mean"),
+        pybind11::arg("invstd"), pybind11::arg("gamma"), pybind11::arg("beta"),
+        pybind11::arg("scale"), pybind11::arg("act"),
+        pybind11::arg("dgamma_acc") = c10::nullopt,
+        pybind11::arg("dbeta_acc") = c10::nullopt);
   m.def("channel_sum_nhwc", &channel_sum_nhwc,
-        "NHWC per-channel sum, fp32 out (gfx950)");
+        "NHWC per-channel sum, fp32 out (gfx950, deterministic)",
+        pybind11::arg("x"), pybind11::arg("acc") = c10::nullopt);
   m.def("conv2d_nhwc_wgrad", &conv2d_nhwc_wgrad,
-        "NHWC wgrad, split-K over pixels, fp32 workspace (gfx950 MFMA)");
+        "NHWC wgrad, split-K over pixel slabs (gfx950 MFMA, deterministic); "
+        "acc: fp32 dense (B,R,S,A)-layout tensor to accumulate into",
+        pybind11::arg("Y"), pybind11::arg("X"), pybind11::arg("R"),
+        pybind11::arg("S"), pybind11::arg("stride"), pybind11::arg("pad"),
+        pybind11::arg("splitp") = 0, pybind11::arg("acc") = c10::nullopt);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC 2x2/s2 maxpool fwd (gfx950)");
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC 2x2/s2 maxpool bwd (gfx950)");
   m.def("upsample2x_fwd", &upsample2x_fwd, "NHWC nearest x2 fwd (gfx950)");

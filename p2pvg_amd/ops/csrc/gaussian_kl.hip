@@ -14,7 +14,7 @@ __global__ void gaussian_kl_fwd_kernel(const float* __restrict__ mu1,
                                        const float* __restrict__ lv1,
                                        const float* __restrict__ mu2,
                                        const float* __restrict__ lv2,
-                                       float* __restrict__ out,  // single scalar
+                                       float* __restrict__ rows,  // per-block
                                        float inv_denom, long n) {
   float acc = 0.f;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -23,11 +23,29 @@ __global__ void gaussian_kl_fwd_kernel(const float* __restrict__ mu1,
     acc += 0.5f * (lv2[i] - lv1[i]) +
            (__expf(lv1[i]) + dmu * dmu) / (2.f * __expf(lv2[i])) - 0.5f;
   }
-  // wave reduce then one atomic per wave
+  // deterministic: wave shfl reduce -> per-wave LDS slot -> serial combine
+  // by thread 0 -> per-block row store (scalar_rows_sum finishes serially)
 #pragma unroll
   for (int off = WAVE / 2; off > 0; off >>= 1)
     acc += __shfl_down(acc, off, WAVE);
-  if ((threadIdx.x & (WAVE - 1)) == 0) atomicAdd(out, acc * inv_denom);
+  __shared__ float sw[256 / WAVE];
+  if ((threadIdx.x & (WAVE - 1)) == 0) sw[threadIdx.x / WAVE] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+#pragma unroll
+    for (int w = 0; w < 256 / WAVE; ++w) t += sw[w];
+    rows[blockIdx.x] = t * inv_denom;
+  }
+}
+
+__global__ void scalar_rows_sum_kernel(const float* __restrict__ rows, int n,
+                                       float* __restrict__ out) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    float t = 0.f;
+    for (int i = 0; i < n; ++i) t += rows[i];
+    out[0] = t;
+  }
 }
 
 __global__ void gaussian_kl_bwd_kernel(
@@ -63,16 +81,20 @@ torch::Tensor gaussian_kl_fwd(torch::Tensor mu1, torch::Tensor lv1,
   CHECK_INPUT(lv1);
   CHECK_INPUT(mu2);
   CHECK_INPUT(lv2);
-  auto out = torch::zeros({}, mu1.options());
+  auto out = torch::empty({}, mu1.options());
   const long n = mu1.numel();
   const int threads = 256;
-  const int blocks = std::min<long>(64, (n + threads - 1) / threads);
+  const int blocks =
+      (int)std::max<long>(1, std::min<long>(64, (n + threads - 1) / threads));
+  auto rows = torch::empty({blocks}, mu1.options());
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(gaussian_kl_fwd_kernel, dim3(std::max(blocks, 1)),
+  hipLaunchKernelGGL(gaussian_kl_fwd_kernel, dim3(blocks),
                      dim3(threads), 0, stream, mu1.data_ptr<float>(),
                      lv1.data_ptr<float>(), mu2.data_ptr<float>(),
-                     lv2.data_ptr<float>(), out.data_ptr<float>(),
+                     lv2.data_ptr<float>(), rows.data_ptr<float>(),
                      (float)(1.0 / denom), n);
+  hipLaunchKernelGGL(scalar_rows_sum_kernel, dim3(1), dim3(64), 0, stream,
+                     rows.data_ptr<float>(), blocks, out.data_ptr<float>());
   return out;
 }
 
@@ -134,7 +156,16 @@ __global__ __launch_bounds__(256) void sqdiff_sum_kernel(
     float t = 0.f;
 #pragma unroll
     for (int w = 0; w < 256 / WAVE; ++w) t += sw[w];
-    atomicAdd(out, t);
+    out[blockIdx.x] = t;   // per-block row; serial combine follows
+  }
+}
+
+__global__ void sq_rows_sum_kernel(const float* __restrict__ rows, int n,
+                                   float* __restrict__ out) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    float t = 0.f;
+    for (int i = 0; i < n; ++i) t += rows[i];
+    out[0] = t;
   }
 }
 
@@ -149,23 +180,26 @@ torch::Tensor sqdiff_sum(torch::Tensor a, torch::Tensor b) {
               a.is_non_overlapping_and_dense(),
               "sqdiff_sum: layouts must match and be dense");
   TORCH_CHECK(a.numel() % 8 == 0, "sqdiff_sum: numel % 8 != 0");
-  auto out = torch::zeros({}, a.options().dtype(torch::kFloat32));
+  auto out = torch::empty({}, a.options().dtype(torch::kFloat32));
   const long nvec = a.numel() / 8;
-  const int grid = (int)std::min<long>(1280, (nvec + 255) / 256);
+  const int grid = (int)std::max<long>(1, std::min<long>(1280, (nvec + 255) / 256));
+  auto rows = torch::empty({grid}, a.options().dtype(torch::kFloat32));
   auto stream = at::cuda::getCurrentCUDAStream();
   if (b.scalar_type() == torch::kBFloat16) {
     hipLaunchKernelGGL((sqdiff_sum_kernel<bf16v8, __bf16>), dim3(grid),
                        dim3(256), 0, stream,
                        reinterpret_cast<const __bf16*>(a.data_ptr()),
                        reinterpret_cast<const __bf16*>(b.data_ptr()),
-                       out.data_ptr<float>(), nvec);
+                       rows.data_ptr<float>(), nvec);
   } else {
     TORCH_CHECK(b.scalar_type() == torch::kFloat32,
                 "sqdiff_sum: b must be bf16 or fp32");
     hipLaunchKernelGGL((sqdiff_sum_kernel<f32v8, float>), dim3(grid),
                        dim3(256), 0, stream,
                        reinterpret_cast<const __bf16*>(a.data_ptr()),
-                       b.data_ptr<float>(), out.data_ptr<float>(), nvec);
+                       b.data_ptr<float>(), rows.data_ptr<float>(), nvec);
   }
+  hipLaunchKernelGGL(sq_rows_sum_kernel, dim3(1), dim3(64), 0, stream,
+                     rows.data_ptr<float>(), grid, out.data_ptr<float>());
   return out;
 }

@@ -150,7 +150,7 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
     const __bf16* __restrict__ x, const __bf16* __restrict__ dy,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
-    float* __restrict__ red,  // (2,K)
+    float* __restrict__ rows,  // (gridDim.x, 2, C) per-block partials (stores)
     long nvec, int C, int act) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
@@ -180,13 +180,14 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
     }
   }
 
-  // three-level combine so the global traffic is ONE atomic per channel per
-  // block: (1) shfl-reduce lanes sharing a channel group within the wave,
-  // (2) wave leaders combine in LDS, (3) one writer pass to global.
+  // deterministic three-level combine, no atomics anywhere:
+  // (1) shfl-reduce lanes sharing a channel group within the wave (fixed
+  // order), (2) each wave STORES its totals into its own LDS slab, (3) a
+  // serial walk over the slabs writes this block's partial ROW to global;
+  // the serial row walk in bn_red_combine finishes the reduction.
   const int lane = threadIdx.x & 63;
-  extern __shared__ float sred[];  // 2*C floats
-  for (int i = threadIdx.x; i < 2 * C; i += BLOCK) sred[i] = 0.f;
-  __syncthreads();
+  extern __shared__ float sred[];  // (BLOCK/64) * 2*C floats
+  const int wave = threadIdx.x >> 6;
   if (cvec < 64) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -199,13 +200,39 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
   if (lane < min(cvec, 64)) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&sred[c0 + j], p1[j]);
-      atomicAdd(&sred[C + c0 + j], p2[j]);
+      sred[wave * 2 * C + c0 + j] = p1[j];
+      sred[wave * 2 * C + C + c0 + j] = p2[j];
     }
   }
   __syncthreads();
+  float* row = rows + (long)blockIdx.x * 2 * C;
   for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-    atomicAdd(&red[i], sred[i]);
+    float t = 0.f;
+#pragma unroll
+    for (int w = 0; w < BLOCK / 64; ++w) t += sred[w * 2 * C + i];
+    row[i] = t;
+  }
+}
+
+// red_final[i] = sum over rows (serial -> deterministic); optionally
+// accumulates dgamma (= s2) / dbeta (= s1) into caller-owned fp32 buffers
+// (the BN params' .grad) so autograd-side accumulation adds disappear.
+__global__ void bn_red_combine_kernel(const float* __restrict__ rows,
+                                      int nrows, float* __restrict__ red_final,
+                                      float* __restrict__ dgamma,
+                                      float* __restrict__ dbeta, int C,
+                                      int accumulate) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < 2 * C;
+       i += gridDim.x * blockDim.x) {
+    float t = 0.f;
+    for (int r = 0; r < nrows; ++r) t += rows[(long)r * 2 * C + i];
+    red_final[i] = t;
+    if (i < C) {
+      if (dbeta != nullptr) dbeta[i] = accumulate ? dbeta[i] + t : t;
+    } else {
+      if (dgamma != nullptr)
+        dgamma[i - C] = accumulate ? dgamma[i - C] + t : t;
+    }
   }
 }
 
@@ -214,7 +241,7 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
 // BN reduce above; ATen's strided (0,2,3) reduction on channels_last runs
 // ~8x off bandwidth on these shapes.
 __global__ __launch_bounds__(BLOCK) void channel_sum_kernel(
-    const __bf16* __restrict__ x, float* __restrict__ out, long nvec, int C) {
+    const __bf16* __restrict__ x, float* __restrict__ rows, long nvec, int C) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float p[8];
@@ -227,9 +254,8 @@ __global__ __launch_bounds__(BLOCK) void channel_sum_kernel(
     for (int j = 0; j < 8; ++j) p[j] += (float)v[j];
   }
   const int lane = threadIdx.x & 63;
-  extern __shared__ float sred[];  // C floats
-  for (int i = threadIdx.x; i < C; i += BLOCK) sred[i] = 0.f;
-  __syncthreads();
+  extern __shared__ float sred[];  // (BLOCK/64) * C floats
+  const int wave = threadIdx.x >> 6;
   if (cvec < 64) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -240,10 +266,27 @@ __global__ __launch_bounds__(BLOCK) void channel_sum_kernel(
   }
   if (lane < min(cvec, 64)) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&sred[c0 + j], p[j]);
+    for (int j = 0; j < 8; ++j) sred[wave * C + c0 + j] = p[j];
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < C; i += BLOCK) atomicAdd(&out[i], sred[i]);
+  for (int i = threadIdx.x; i < C; i += BLOCK) {
+    float t = 0.f;
+#pragma unroll
+    for (int w = 0; w < BLOCK / 64; ++w) t += sred[w * C + i];
+    rows[(long)blockIdx.x * C + i] = t;
+  }
+}
+
+// out[c] (+)= serial sum over per-block partial rows (deterministic).
+__global__ void rows_combine_kernel(const float* __restrict__ rows, int nrows,
+                                    float* __restrict__ out, long C,
+                                    int accumulate) {
+  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += (long)gridDim.x * blockDim.x) {
+    float t = 0.f;
+    for (int r = 0; r < nrows; ++r) t += rows[(long)r * C + c];
+    out[c] = accumulate ? out[c] + t : t;
+  }
 }
 
 // Small-C channel sum (C <= 4: the decoder's nc-channel output, where ATen's
@@ -251,7 +294,7 @@ __global__ __launch_bounds__(BLOCK) void channel_sum_kernel(
 // C fp32 register partials, full-wave shfl reduce, one atomic per block.
 template <int C>
 __global__ __launch_bounds__(BLOCK) void channel_sum_smallc_kernel(
-    const __bf16* __restrict__ x, float* __restrict__ out, long npix) {
+    const __bf16* __restrict__ x, float* __restrict__ rows, long npix) {
   float p[C];
 #pragma unroll
   for (int c = 0; c < C; ++c) p[c] = 0.f;
@@ -279,7 +322,7 @@ __global__ __launch_bounds__(BLOCK) void channel_sum_smallc_kernel(
       float t = 0.f;
 #pragma unroll
       for (int w = 0; w < BLOCK / 64; ++w) t += sw[w][c];
-      atomicAdd(&out[c], t);
+      rows[(long)blockIdx.x * C + c] = t;
     }
   }
 }
@@ -395,25 +438,40 @@ torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
 }
 
 // Backward: returns (dx, dgamma, dbeta). Reads only x and dy (activation
-// derivative recomputed from the pre-activation value).
+// derivative recomputed from the pre-activation value). Deterministic by
+// construction (per-block partial rows + serial combine). When
+// dgamma_acc/dbeta_acc are given (the BN params' .grad), the combine
+// ACCUMULATES into them in place and returns them.
 std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                                       torch::Tensor mean, torch::Tensor invstd,
                                       torch::Tensor gamma, torch::Tensor beta,
-                                      torch::Tensor scale, long act) {
+                                      torch::Tensor scale, long act,
+                                      c10::optional<torch::Tensor> dgamma_acc,
+                                      c10::optional<torch::Tensor> dbeta_acc) {
   const int C = x.size(1);
   const long count = x.numel() / C;
   auto f32 = x.options().dtype(torch::kFloat32);
-  auto red = torch::zeros({2, C}, f32);
   auto dx = torch::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   const long nvec = x.numel() / 8;
-  hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(pick_grid(nvec)),
-                     dim3(BLOCK), 2 * C * sizeof(float), stream,
+  const int rgrid = pick_grid(nvec);
+  auto rows = torch::empty({rgrid, 2, C}, f32);     // stores — never zeroed
+  auto red = torch::empty({2, C}, f32);
+  hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(rgrid), dim3(BLOCK),
+                     (BLOCK / 64) * 2 * C * sizeof(float), stream,
                      reinterpret_cast<const __bf16*>(x.data_ptr()),
                      reinterpret_cast<const __bf16*>(dy.data_ptr()),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                     red.data_ptr<float>(), nvec, C, (int)act);
+                     rows.data_ptr<float>(), nvec, C, (int)act);
+  const int accumulate = dgamma_acc.has_value() ? 1 : 0;
+  torch::Tensor dgamma =
+      accumulate ? dgamma_acc.value() : torch::empty({C}, f32);
+  torch::Tensor dbeta = accumulate ? dbeta_acc.value() : torch::empty({C}, f32);
+  hipLaunchKernelGGL(bn_red_combine_kernel, dim3(ceil_div(2 * C, 256)),
+                     dim3(256), 0, stream, rows.data_ptr<float>(), rgrid,
+                     red.data_ptr<float>(), dgamma.data_ptr<float>(),
+                     dbeta.data_ptr<float>(), C, accumulate);
   hipLaunchKernelGGL(bn_act_bwd_apply_kernel, dim3(pick_grid(nvec)),
                      dim3(BLOCK), 0, stream,
                      reinterpret_cast<const __bf16*>(x.data_ptr()),
@@ -423,26 +481,31 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                      scale.data_ptr<float>(), red.data_ptr<float>(),
                      reinterpret_cast<__bf16*>(dx.data_ptr()), nvec, C,
                      (int)act, (float)(1.0 / count));
-  // dgamma = s2 (sum dy'*xhat), dbeta = s1
-  auto dgamma = red[1].clone();
-  auto dbeta = red[0].clone();
   return {dx, dgamma, dbeta};
 }
 
 // db = per-channel sum of a channels_last bf16 tensor, fp32 out.
-torch::Tensor channel_sum_nhwc(torch::Tensor x) {
+// Deterministic (row stores + serial combine). With `acc` (the bias .grad),
+// accumulates into it in place and returns it.
+torch::Tensor channel_sum_nhwc(torch::Tensor x,
+                               c10::optional<torch::Tensor> acc) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
               x.is_contiguous(at::MemoryFormat::ChannelsLast),
               "channel_sum_nhwc: need channels_last bf16 CUDA tensor");
   const int C = x.size(1);
-  auto out = torch::zeros({C}, x.options().dtype(torch::kFloat32));
+  auto f32 = x.options().dtype(torch::kFloat32);
+  const int accumulate = acc.has_value() ? 1 : 0;
+  torch::Tensor out = accumulate ? acc.value() : torch::empty({C}, f32);
   auto stream = at::cuda::getCurrentCUDAStream();
+  int grid;
+  torch::Tensor rows;
   if (C <= 4) {
     // decoder output channels (nc = 1 or 3): pixel-major scalar walk
     const long npix = x.numel() / C;
-    const int grid = (int)std::min<long>(2048, (npix + BLOCK - 1) / BLOCK);
+    grid = (int)std::min<long>(2048, (npix + BLOCK - 1) / BLOCK);
+    rows = torch::empty({grid, C}, f32);
     const __bf16* px = reinterpret_cast<const __bf16*>(x.data_ptr());
-    float* po = out.data_ptr<float>();
+    float* po = rows.data_ptr<float>();
     switch (C) {
       case 1: hipLaunchKernelGGL((channel_sum_smallc_kernel<1>), dim3(grid),
                                  dim3(BLOCK), 0, stream, px, po, npix); break;
@@ -453,14 +516,20 @@ torch::Tensor channel_sum_nhwc(torch::Tensor x) {
       default: hipLaunchKernelGGL((channel_sum_smallc_kernel<4>), dim3(grid),
                                   dim3(BLOCK), 0, stream, px, po, npix);
     }
-    return out;
+  } else {
+    TORCH_CHECK(C % 8 == 0 && 256 % (C / 8) == 0,
+                "channel_sum_nhwc: C/8 must divide 256 (or C <= 4)");
+    const long nvec = x.numel() / 8;
+    grid = pick_grid(nvec);
+    rows = torch::empty({grid, C}, f32);
+    hipLaunchKernelGGL(channel_sum_kernel, dim3(grid), dim3(BLOCK),
+                       (BLOCK / 64) * C * sizeof(float), stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       rows.data_ptr<float>(), nvec, C);
   }
-  TORCH_CHECK(C % 8 == 0 && 256 % (C / 8) == 0,
-              "channel_sum_nhwc: C/8 must divide 256 (or C <= 4)");
-  const long nvec = x.numel() / 8;
-  hipLaunchKernelGGL(channel_sum_kernel, dim3(pick_grid(nvec)), dim3(BLOCK),
-                     C * sizeof(float), stream,
-                     reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     out.data_ptr<float>(), nvec, C);
+  hipLaunchKernelGGL(rows_combine_kernel,
+                     dim3((int)std::min(32L, (long)ceil_div(C, 256))),
+                     dim3(256), 0, stream, rows.data_ptr<float>(), grid,
+                     out.data_ptr<float>(), (long)C, accumulate);
   return out;
 }

@@ -52,15 +52,32 @@ class FusedBNActFn(torch.autograd.Function):
         )
         ctx.save_for_backward(x, mean, invstd, gamma32, beta32, scale)
         ctx.act = act
+        ctx.gref, ctx.bref = gamma, beta
         return y
 
     @staticmethod
     def backward(ctx, dy):
+        from .conv import _acc_target, weight_grads_enabled
+
         x, mean, invstd, gamma32, beta32, scale = ctx.saved_tensors
         ext = _ext()
         dy = dy.contiguous(memory_format=CL)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
+        if not weight_grads_enabled():
+            # phase-2 traversal: dx only, param grads are discarded anyway
+            dx, _, _ = ext.bn_act_bwd(
+                x, dy, mean, invstd, gamma32, beta32, scale, ctx.act
+            )
+            return dx, None, None, None, None, None, None, None, None
+        ga = _acc_target(ctx.gref)
+        ba = _acc_target(ctx.bref)
+        if ga is not None and ba is not None:
+            # accumulate dgamma/dbeta straight into the managed .grad buffers
+            dx, _, _ = ext.bn_act_bwd(
+                x, dy, mean, invstd, gamma32, beta32, scale, ctx.act, ga, ba
+            )
+            return dx, None, None, None, None, None, None, None, None
         dx, dgamma, dbeta = ext.bn_act_bwd(
             x, dy, mean, invstd, gamma32, beta32, scale, ctx.act
         )

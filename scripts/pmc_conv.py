@@ -15,6 +15,6 @@ x = (torch.randn(N, C, H, W, device="cuda") * 0.5).bfloat16().contiguous(memory_
 w = (torch.randn(K, C, 3, 3, device="cuda") * 0.02).bfloat16().contiguous(memory_format=CL)
 b = torch.randn(K, device="cuda").float()
 for _ in range(10):
-    out = ext.conv2d_nhwc_fwd(x, w, b, 1, 1, 0, None)
+    out = ext.conv2d_nhwc_fwd(x, w, b, 1, 1, 0, False)[0]
 torch.cuda.synchronize()
 print("ok", out.shape)

@@ -55,7 +55,7 @@ def test_conv_fwd_matches_aten(ext, N, C, H, W, K, ks, st, pad, act):
 
     xl = x.bfloat16().contiguous(memory_format=torch.channels_last)
     wl = w.bfloat16().contiguous(memory_format=torch.channels_last)
-    out = ext.conv2d_nhwc_fwd(xl, wl, b, st, pad, act)
+    out = ext.conv2d_nhwc_fwd(xl, wl, b, st, pad, act)[0]
 
     assert out.shape == ref.shape
     out_f = out.float()
@@ -78,7 +78,7 @@ def test_conv_fwd_perf_vs_miopen(ext):
     wl = w.contiguous(memory_format=torch.channels_last)
 
     def ours():
-        return ext.conv2d_nhwc_fwd(xl, wl, b, 1, 1, 0)
+        return ext.conv2d_nhwc_fwd(xl, wl, b, 1, 1, 0)[0]
 
     def theirs():
         return torch.nn.functional.conv2d(xl, wl, b.bfloat16(), stride=1, padding=1)

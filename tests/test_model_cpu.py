@@ -161,3 +161,23 @@ def test_conv_weight_shadows_track_weight_version():
         sht["f"].float(), ct.weight.detach().bfloat16().transpose(0, 1).float()
     )
     torch.testing.assert_close(sht["b"].float(), ct.weight.detach().bfloat16().float())
+
+
+def test_zero_grads_stable_buffers(tiny_cfg):
+    """zero_grads materializes every .grad once and keeps buffer identity
+    stable across steps (hipGraph capture + fused Adam rely on this)."""
+    import torch
+
+    from p2pvg_amd.models import P2PModel
+
+    model = P2PModel(tiny_cfg)
+    model.zero_grads()
+    ids = {n: id(p.grad) for n, p in model.named_parameters()}
+    assert all(p.grad is not None for p in model.parameters())
+    x = torch.rand(4, tiny_cfg.batch_size, tiny_cfg.channels,
+                   tiny_cfg.image_width, tiny_cfg.image_width)
+    model(x, 0, 3)
+    model.zero_grads()
+    for n, p in model.named_parameters():
+        assert id(p.grad) == ids[n], f"{n}: grad buffer identity changed"
+        assert float(p.grad.abs().max()) == 0.0
