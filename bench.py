@@ -84,8 +84,11 @@ def main():
         backend = "nccl" if use_cuda else "gloo"
         dist.init_process_group(backend=backend)
     if use_cuda:
-        torch.cuda.set_device(local_rank)
-        device = torch.device(f"cuda:{local_rank}")
+        # oversubscription probe: N ranks on fewer GPUs (e.g. 2-rank RCCL on
+        # one MI355X to exercise real collectives within a 1-GPU lease)
+        dev_idx = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = torch.device(f"cuda:{dev_idx}")
     else:
         device = torch.device("cpu")
 

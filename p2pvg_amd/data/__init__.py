@@ -107,9 +107,12 @@ def get_generator(loader: DataLoader, device, dynamic_length: bool = True) -> It
                 # permute().contiguous() would allocate an unpinned
                 # intermediate and silently degrade the copy to synchronous),
                 # then do the (b,t)->(t,b) transpose on device where it costs
-                # one ~30us pass at HBM bandwidth
+                # one ~30us pass at HBM bandwidth. Store the clip NHWC so
+                # each x[i] is a channels_last (B,C,H,W) frame — the NHWC
+                # conv path then never re-lays frames out per encoder call.
                 data = data.to(device, non_blocking=True)
-                data = data.permute(1, 0, 2, 3, 4).contiguous()
+                data = (data.permute(1, 0, 3, 4, 2).contiguous()
+                        .permute(0, 1, 4, 2, 3))
             else:
                 data = data.permute(1, 0, 2, 3, 4).contiguous().to(device)
             if dynamic_length:

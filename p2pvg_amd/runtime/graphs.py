@@ -44,13 +44,46 @@ class _GraphEntry:
 
 class GraphedTrainStep:
     def __init__(self, model, amp_dtype: Optional[torch.dtype] = None,
-                 warmup_iters: int = 3, max_graphs: int = 16):
+                 warmup_iters: int = 1, max_graphs: int = 16):
         self.model = model
         self.amp_dtype = amp_dtype
         self.warmup_iters = warmup_iters
         self.max_graphs = max_graphs
         self.graphs: Dict[Tuple, _GraphEntry] = {}
         self.eager_keys = set()
+        self._state_ready = False
+
+    def _materialize_persistent_state(self):
+        """Allocate every step-persistent tensor (grad buffers, Adam state,
+        conv weight shadows) BEFORE any warmup runs. Without this, the first
+        warmup interleaves persistent allocations with a full step's
+        transient activations, and the resulting fragmentation (persistent
+        tensors pinning partially-used segments) was the vgg_128 batch-128
+        capture OOM: 116 GB reserved-but-unallocated that the graph's
+        private pool cannot reuse."""
+        if self._state_ready:
+            return
+        model = self.model
+        model.zero_grads()
+        from ..optim import HIPFusedAdam
+
+        for _, opt in self._optimizers():
+            if isinstance(opt, HIPFusedAdam):
+                for group in opt.param_groups:
+                    for p in group["params"]:
+                        state = opt.state[p]
+                        if len(state) == 0:
+                            state["step"] = torch.zeros(
+                                (), dtype=torch.float32, device=p.device)
+                            state["exp_avg"] = torch.zeros_like(p)
+                            state["exp_avg_sq"] = torch.zeros_like(p)
+        try:
+            from ..ops.conv import refresh_conv_shadows
+
+            refresh_conv_shadows(model)
+        except Exception:  # noqa: BLE001 — shadows only exist on the HIP path
+            pass
+        self._state_ready = True
 
     def _amp_ctx(self, cache_enabled: bool):
         if self.amp_dtype is not None:
@@ -128,6 +161,7 @@ class GraphedTrainStep:
         entry.dts = dts.clone()
         entry.loss_out = torch.zeros(4, device=prev.device)
 
+        self._materialize_persistent_state()
         snap = self._snapshot_train_state()
         # warmup on a side stream (standard CUDAGraph recipe)
         s = torch.cuda.Stream()
@@ -175,7 +209,7 @@ class GraphedTrainStep:
             try:
                 entry = self._capture(plan, prev, cur, tun, dts)
                 self.graphs[key] = entry
-            except torch.OutOfMemoryError:
+            except (torch.OutOfMemoryError, RuntimeError):
                 # capture needs headroom beyond the eager peak (side-stream
                 # warmup segments + the graph's private pool cannot share
                 # the allocator's fragmented free blocks, and

@@ -128,6 +128,10 @@ def main():
         )
 
     model = P2PModel(cfg).to(device)
+    if device.type == "cuda" and cfg.dataset != "h36m":
+        # NHWC params/activations: the gfx950 conv kernels are NHWC-native
+        # (and MIOpen igemm too on the fallback path)
+        model = model.to(memory_format=torch.channels_last)
 
     start_epoch = 0
     if cfg.ckpt:
@@ -146,13 +150,19 @@ def main():
 
     stepper = None
     if cfg.use_graphs and device.type == "cuda":
-        if world > 1:
-            logger.info("[!] hipGraph capture with RCCL collectives is untested; "
-                        "running DDP eager (graphs disabled)")
-        else:
-            from p2pvg_amd.runtime import GraphedTrainStep
+        # With DDP the two RCCL all-reduce flush points are recorded INSIDE
+        # the captured step (grad_sync runs in _backward_and_step). RCCL
+        # supports stream-capture of collectives; if capture fails on any
+        # rank the stepper falls back to the eager step for that shape key,
+        # so --use_graphs is safe to combine with --ddp (measured eager vs
+        # graphed numbers: docs/ROADMAP.md).
+        from p2pvg_amd.runtime import GraphedTrainStep
 
-            stepper = GraphedTrainStep(model, amp_dtype=amp_dtype)
+        if world > 1:
+            logger.info("[*] hipGraph capture with RCCL collectives recorded "
+                        "in-graph (falls back to eager per shape key on "
+                        "capture failure)")
+        stepper = GraphedTrainStep(model, amp_dtype=amp_dtype)
 
     for epoch in range(start_epoch, cfg.nepochs):
         model.train()
