@@ -183,6 +183,9 @@ def main():
     if use_cuda:
         import sys
 
+        if cfg.use_graphs:
+            print(f"[bench] rank {rank} graphs: {len(stepper.graphs)} captured, "
+                  f"{len(stepper.eager_keys)} eager keys", file=sys.stderr)
         print(
             f"[bench] rank {rank} peak GB: "
             f"allocated {torch.cuda.max_memory_allocated() / 2**30:.1f} "

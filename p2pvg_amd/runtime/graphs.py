@@ -209,7 +209,13 @@ class GraphedTrainStep:
             try:
                 entry = self._capture(plan, prev, cur, tun, dts)
                 self.graphs[key] = entry
-            except (torch.OutOfMemoryError, RuntimeError):
+            except (torch.OutOfMemoryError, RuntimeError) as e:
+                import sys
+                import traceback
+
+                print(f"[graphs] capture failed for key {key}: {e!r} — "
+                      "running this shape eagerly", file=sys.stderr)
+                traceback.print_exc(file=sys.stderr)
                 # capture needs headroom beyond the eager peak (side-stream
                 # warmup segments + the graph's private pool cannot share
                 # the allocator's fragmented free blocks, and
