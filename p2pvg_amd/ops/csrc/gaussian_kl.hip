@@ -39,13 +39,21 @@ __global__ void gaussian_kl_fwd_kernel(const float* __restrict__ mu1,
   }
 }
 
+// fixed-structure (deterministic) 256-thread strided sum + LDS tree; the
+// single-thread serial walk cost ~59 us at n=1280 from the dependent-load
+// chain — this is ~3 us.
 __global__ void scalar_rows_sum_kernel(const float* __restrict__ rows, int n,
                                        float* __restrict__ out) {
-  if (blockIdx.x == 0 && threadIdx.x == 0) {
-    float t = 0.f;
-    for (int i = 0; i < n; ++i) t += rows[i];
-    out[0] = t;
+  float t = 0.f;
+  for (int i = threadIdx.x; i < n; i += 256) t += rows[i];
+  __shared__ float sw[256];
+  sw[threadIdx.x] = t;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if ((int)threadIdx.x < off) sw[threadIdx.x] += sw[threadIdx.x + off];
+    __syncthreads();
   }
+  if (threadIdx.x == 0) out[0] = sw[0];
 }
 
 __global__ void gaussian_kl_bwd_kernel(
@@ -93,7 +101,7 @@ torch::Tensor gaussian_kl_fwd(torch::Tensor mu1, torch::Tensor lv1,
                      lv1.data_ptr<float>(), mu2.data_ptr<float>(),
                      lv2.data_ptr<float>(), rows.data_ptr<float>(),
                      (float)(1.0 / denom), n);
-  hipLaunchKernelGGL(scalar_rows_sum_kernel, dim3(1), dim3(64), 0, stream,
+  hipLaunchKernelGGL(scalar_rows_sum_kernel, dim3(1), dim3(256), 0, stream,
                      rows.data_ptr<float>(), blocks, out.data_ptr<float>());
   return out;
 }
@@ -162,11 +170,16 @@ __global__ __launch_bounds__(256) void sqdiff_sum_kernel(
 
 __global__ void sq_rows_sum_kernel(const float* __restrict__ rows, int n,
                                    float* __restrict__ out) {
-  if (blockIdx.x == 0 && threadIdx.x == 0) {
-    float t = 0.f;
-    for (int i = 0; i < n; ++i) t += rows[i];
-    out[0] = t;
+  float t = 0.f;
+  for (int i = threadIdx.x; i < n; i += 256) t += rows[i];
+  __shared__ float sw[256];
+  sw[threadIdx.x] = t;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if ((int)threadIdx.x < off) sw[threadIdx.x] += sw[threadIdx.x + off];
+    __syncthreads();
   }
+  if (threadIdx.x == 0) out[0] = sw[0];
 }
 
 }  // namespace
@@ -199,7 +212,7 @@ torch::Tensor sqdiff_sum(torch::Tensor a, torch::Tensor b) {
                        reinterpret_cast<const __bf16*>(a.data_ptr()),
                        b.data_ptr<float>(), rows.data_ptr<float>(), nvec);
   }
-  hipLaunchKernelGGL(sq_rows_sum_kernel, dim3(1), dim3(64), 0, stream,
+  hipLaunchKernelGGL(sq_rows_sum_kernel, dim3(1), dim3(256), 0, stream,
                      rows.data_ptr<float>(), grid, out.data_ptr<float>());
   return out;
 }

@@ -368,6 +368,39 @@ int pick_grid(long nvec) {
   return (int)std::min<long>(2048, (nvec + BLOCK - 1) / BLOCK);
 }
 
+// fold rows [p*64, min(p*64+64, R)) of a (R, C) fp32 matrix into partial row
+// p — serial within the chunk, chunks independent: a fixed-structure
+// deterministic tree level (the serial flat walk was 11 ms at R=28672).
+__global__ void fold_rows_kernel(const float* __restrict__ rows,
+                                 float* __restrict__ out, int R, long C) {
+  const int p = blockIdx.y;
+  const int b0 = p * 64;
+  const int b1 = min(b0 + 64, R);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < C;
+       i += (long)gridDim.x * blockDim.x) {
+    float t = 0.f;
+    for (int b = b0; b < b1; ++b) t += rows[(long)b * C + i];
+    out[(long)p * C + i] = t;
+  }
+}
+
+// deterministically reduce a (R, C) partial matrix to (<=64, C): repeated
+// 64-way folds, each a fixed association order -> bitwise-stable.
+torch::Tensor reduce_rows_det(torch::Tensor rows, long C) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int R = (int)(rows.numel() / C);
+  while (R > 64) {
+    const int R2 = ceil_div(R, 64);
+    auto next = torch::empty({R2, C}, rows.options());
+    dim3 grid((unsigned)std::min<long>(256, ceil_div(C, 256)), (unsigned)R2);
+    hipLaunchKernelGGL(fold_rows_kernel, grid, dim3(256), 0, stream,
+                       rows.data_ptr<float>(), next.data_ptr<float>(), R, C);
+    rows = next;
+    R = R2;
+  }
+  return rows;
+}
+
 }  // namespace
 
 // Training fwd: x conv output (already stats-accumulated if stats given,
@@ -391,7 +424,11 @@ std::vector<torch::Tensor> bn_act_fwd_train(
   auto y = torch::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
 
-  const int nbuckets = stats.dim() == 3 ? (int)stats.size(0) : 1;
+  int nbuckets = stats.dim() == 3 ? (int)stats.size(0) : 1;
+  if (nbuckets > 64) {
+    stats = reduce_rows_det(stats, 2L * C);
+    nbuckets = (int)stats.size(0);
+  }
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(ceil_div(C, 256)), dim3(256), 0,
                      stream, stats.data_ptr<float>(), gamma.data_ptr<float>(),
                      beta.data_ptr<float>(), mean.data_ptr<float>(),
@@ -468,8 +505,14 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
   torch::Tensor dgamma =
       accumulate ? dgamma_acc.value() : torch::empty({C}, f32);
   torch::Tensor dbeta = accumulate ? dbeta_acc.value() : torch::empty({C}, f32);
+  torch::Tensor folded = rows;
+  int nrows = rgrid;
+  if (nrows > 64) {
+    folded = reduce_rows_det(rows, 2L * C);
+    nrows = (int)folded.size(0);
+  }
   hipLaunchKernelGGL(bn_red_combine_kernel, dim3(ceil_div(2 * C, 256)),
-                     dim3(256), 0, stream, rows.data_ptr<float>(), rgrid,
+                     dim3(256), 0, stream, folded.data_ptr<float>(), nrows,
                      red.data_ptr<float>(), dgamma.data_ptr<float>(),
                      dbeta.data_ptr<float>(), C, accumulate);
   hipLaunchKernelGGL(bn_act_bwd_apply_kernel, dim3(pick_grid(nvec)),
@@ -527,8 +570,12 @@ torch::Tensor channel_sum_nhwc(torch::Tensor x,
                        reinterpret_cast<const __bf16*>(x.data_ptr()),
                        rows.data_ptr<float>(), nvec, C);
   }
+  if (grid > 64) {
+    rows = reduce_rows_det(rows, (long)C);
+    grid = (int)rows.size(0);
+  }
   hipLaunchKernelGGL(rows_combine_kernel,
-                     dim3((int)std::min(32L, (long)ceil_div(C, 256))),
+                     dim3((int)std::max(1L, std::min(32L, (long)ceil_div(C, 256)))),
                      dim3(256), 0, stream, rows.data_ptr<float>(), grid,
                      out.data_ptr<float>(), (long)C, accumulate);
   return out;
