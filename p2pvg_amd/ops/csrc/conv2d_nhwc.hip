@@ -27,10 +27,28 @@
 
 #include "common.h"
 
+#include <cstdlib>
+
+// glds 3-buffer pipeline for the in-bounds (pad=0 / padded-input) case
+std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
+                                           c10::optional<torch::Tensor> bias,
+                                           long stride, long act,
+                                           bool want_stats, long oh, long ow,
+                                           long oy0, long ox0);
+bool conv2d_glds_eligible(long C, long K, long R, long stride, long pad);
+
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
+
+bool glds_enabled() {
+  static int v = [] {
+    const char* e = std::getenv("P2PVG_GLDS");
+    return (e != nullptr && e[0] == '0') ? 0 : 1;
+  }();
+  return v != 0;
+}
 
 constexpr int THREADS = 256;
 
@@ -441,6 +459,11 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
   TORCH_CHECK(R == S, "square kernels only");
   const int HO = (H + 2 * (int)pad - R) / (int)stride + 1;
   const int WO = (W + 2 * (int)pad - R) / (int)stride + 1;
+
+  if (conv2d_glds_eligible(C, K, R, stride, pad) &&
+      (long)Nb * HO * WO >= 4096 && glds_enabled()) {
+    return conv2d_glds_fwd(in, w, bias, stride, act, want_stats, 0, 0, 0, 0);
+  }
 
   auto out = torch::empty({Nb, K, HO, WO},
                           in.options().memory_format(at::MemoryFormat::ChannelsLast));

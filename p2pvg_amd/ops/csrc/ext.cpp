@@ -32,6 +32,11 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
 std::vector<torch::Tensor> conv2d_nhwc_fracstride(
     torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
     long up_stride, long up_pad, long OH, long OW, long act, bool want_stats);
+std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
+                                           c10::optional<torch::Tensor> bias,
+                                           long stride, long act,
+                                           bool want_stats, long oh, long ow,
+                                           long oy0, long ox0);
 std::vector<torch::Tensor> bn_act_fwd_train(
     torch::Tensor x, torch::Tensor stats, torch::Tensor gamma,
     torch::Tensor beta, c10::optional<torch::Tensor> running_mean,
@@ -72,6 +77,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("up_stride"), pybind11::arg("up_pad"), pybind11::arg("OH"),
         pybind11::arg("OW"), pybind11::arg("act"),
         pybind11::arg("want_stats") = false);
+  m.def("conv2d_glds_fwd", &conv2d_glds_fwd,
+        "NHWC conv fwd, glds 3-buffer pipeline (padded-input contract)",
+        pybind11::arg("in"), pybind11::arg("w"), pybind11::arg("bias"),
+        pybind11::arg("stride"), pybind11::arg("act"),
+        pybind11::arg("want_stats") = false, pybind11::arg("oh") = 0,
+        pybind11::arg("ow") = 0, pybind11::arg("oy0") = 0,
+        pybind11::arg("ox0") = 0);
   m.def("bn_act_fwd_train", &bn_act_fwd_train, "fused BN+act train fwd (gfx950)");
   m.def("bn_act_fwd_eval", &bn_act_fwd_eval, "fused BN+act eval fwd (gfx950)");
   m.def("bn_act_bwd", &bn_act_bwd, "fused BN+act bwd (gfx950, deterministic)",

@@ -98,3 +98,65 @@ def test_conv_fwd_perf_vs_miopen(ext):
     print(f"\nconv fwd 128x256x16x16 k3: hip={times['hip']:.1f}us miopen={times['miopen']:.1f}us")
     # sanity floor only: within 4x of MIOpen (tightened as the kernel is tuned)
     assert times["hip"] < 4 * times["miopen"]
+
+
+@pytest.mark.parametrize("n,c,h,k,ks,st", [
+    (4, 64, 32, 64, 3, 1),     # BN=64 path
+    (4, 64, 32, 128, 3, 1),    # BN=128 path
+    (4, 128, 16, 256, 3, 1),
+    (3, 64, 32, 128, 4, 2),    # dcgan k4s2
+    (2, 192, 16, 192, 3, 1),   # C not mult of BN row-group edge
+])
+def test_glds_conv_matches_reference(ext, n, c, h, ks, st, k):
+    """glds 3-buffer kernel on a pre-padded input == fp32 conv with pad=1."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    x = torch.randn(n, c, h, h, device="cuda").bfloat16().contiguous(memory_format=CL)
+    w = (torch.randn(k, c, ks, ks, device="cuda") * 0.05).bfloat16() \
+        .contiguous(memory_format=CL)
+    xp = F.pad(x.float(), (1, 1, 1, 1)).bfloat16().contiguous(memory_format=CL)
+    got = ext.conv2d_glds_fwd(xp, w, None, st, 0, False)[0]
+    ref = F.conv2d(x.float(), w.float(), None, st, 1)
+    assert got.shape == ref.shape
+    err = (got.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 0.02 * scale + 0.02, f"err {err} scale {scale}"
+
+
+def test_glds_conv_stats_and_act(ext):
+    """Fused epilogue: leaky act + per-channel sum/sumsq bucket stores."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(1)
+    n, c, h, k = 4, 64, 32, 128
+    x = torch.randn(n, c, h, h, device="cuda").bfloat16().contiguous(memory_format=CL)
+    w = (torch.randn(k, c, 3, 3, device="cuda") * 0.05).bfloat16() \
+        .contiguous(memory_format=CL)
+    b = torch.randn(k, device="cuda")
+    xp = F.pad(x.float(), (1, 1, 1, 1)).bfloat16().contiguous(memory_format=CL)
+    got, stats = ext.conv2d_glds_fwd(xp, w, b, 1, 1, True)
+    ref = F.leaky_relu(F.conv2d(x.float(), w.float(), b, 1, 1), 0.2)
+    err = (got.float() - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item() + 0.02
+    s = stats.sum(dim=0)
+    want_sum = ref.to(torch.bfloat16).float().sum(dim=(0, 2, 3))
+    assert torch.allclose(s[0], want_sum, rtol=2e-2, atol=2e-1), \
+        (s[0] - want_sum).abs().max()
+
+
+def test_glds_padded_output_placement(ext):
+    """oy0/ox0/OH/OW place the interior inside a larger physical output."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(2)
+    n, c, h, k = 2, 64, 16, 64
+    x = torch.randn(n, c, h, h, device="cuda").bfloat16().contiguous(memory_format=CL)
+    w = (torch.randn(k, c, 3, 3, device="cuda") * 0.05).bfloat16() \
+        .contiguous(memory_format=CL)
+    xp = F.pad(x.float(), (1, 1, 1, 1)).bfloat16().contiguous(memory_format=CL)
+    got = ext.conv2d_glds_fwd(xp, w, None, 1, 0, False, h + 2, h + 2, 1, 1)[0]
+    assert got.shape == (n, k, h + 2, h + 2)
+    ref = F.conv2d(x.float(), w.float(), None, 1, 1)
+    err = (got.float()[:, :, 1:-1, 1:-1] - ref).abs().max().item()
+    assert err < 0.02 * ref.abs().max().item() + 0.02
