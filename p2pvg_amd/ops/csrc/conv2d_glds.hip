@@ -44,7 +44,16 @@ __device__ __forceinline__ float gactivate(float v, int act) {
 
 template <int N>
 __device__ __forceinline__ void waitcnt_vm() {
-  asm volatile("s_waitcnt vmcnt(%0)" ::"n"(N));
+  // "memory" clobber: the raw barrier/wait pair is NOT an IR-level memory
+  // fence, so without it the compiler may hoist the buffer's ds_reads above
+  // the wait (observed as a sporadic large-shape numerics failure)
+  asm volatile("s_waitcnt vmcnt(%0)" ::"n"(N) : "memory");
+}
+
+__device__ __forceinline__ void barrier_mem() {
+  asm volatile("" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
 }
 
 __device__ __forceinline__ void glds16(const __bf16* g, char* l) {
@@ -179,7 +188,7 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
     const char* bbuf = abuf + ABYTES;
     // chunk t landed once <= 2 chunks' glds remain outstanding
     waitcnt_vm<2 * G>();
-    __builtin_amdgcn_s_barrier();
+    barrier_mem();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int cb = kk * 64 + ((lane >> 4) * 16);
@@ -201,7 +210,7 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_barrier();       // everyone done reading buf
+    barrier_mem();                      // everyone done reading buf
     if (t + 3 < nchunks) stage(t + 3, buf);
   }
 

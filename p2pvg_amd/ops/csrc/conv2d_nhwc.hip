@@ -461,7 +461,8 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
   const int WO = (W + 2 * (int)pad - R) / (int)stride + 1;
 
   if (conv2d_glds_eligible(C, K, R, stride, pad) &&
-      (long)Nb * HO * WO >= 4096 && glds_enabled()) {
+      ((long)Nb * HO * WO / 256) * (K / (K % 128 ? 64 : 128)) >= 200 &&
+      glds_enabled()) {
     return conv2d_glds_fwd(in, w, bias, stride, act, want_stats, 0, 0, 0, 0);
   }
 
