@@ -18,14 +18,18 @@ from ...ops.fused_norm import FusedSequential
 
 
 class dcgan_conv(nn.Module):
-    """Conv2d(k4,s2,p1) + BatchNorm2d + LeakyReLU(0.2)."""
+    """Conv2d(k4,s2,p1) + BatchNorm2d + LeakyReLU(0.2).
 
-    def __init__(self, nin: int, nout: int):
+    pad_out: emit the activation as a padded zero-ring map so the NEXT
+    stride-2 conv's gathers are in-bounds (glds staging pipeline)."""
+
+    def __init__(self, nin: int, nout: int, pad_out: bool = False):
         super().__init__()
         self.main = FusedSequential(
             Conv2d(nin, nout, 4, 2, 1),
             BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
+            pad_out=pad_out,
         )
 
     def forward(self, x):
@@ -48,14 +52,19 @@ class dcgan_upconv(nn.Module):
 
 
 class vgg_layer(nn.Module):
-    """Conv2d(k3,s1,p1) + BatchNorm2d + LeakyReLU(0.2)."""
+    """Conv2d(k3,s1,p1) + BatchNorm2d + LeakyReLU(0.2).
 
-    def __init__(self, nin: int, nout: int):
+    pad_out: emit the activation as a padded zero-ring map for the consumer
+    OUTSIDE this block (pool, upsample, or the next stage's conv); inside a
+    FusedSequential chain the ring decision is automatic."""
+
+    def __init__(self, nin: int, nout: int, pad_out: bool = False):
         super().__init__()
         self.main = FusedSequential(
             Conv2d(nin, nout, 3, 1, 1),
             BatchNorm2d(nout),
             nn.LeakyReLU(0.2, inplace=True),
+            pad_out=pad_out,
         )
 
     def forward(self, x):

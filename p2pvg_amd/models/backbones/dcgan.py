@@ -17,15 +17,24 @@ from ...ops.fused_norm import FusedSequential
 from .blocks import dcgan_conv, dcgan_upconv
 
 
+def _interior(t):
+    """Logical (interior) view of a padded zero-ring map; identity on dense
+    tensors. The DCGAN decoder runs dense, so padded encoder skips are
+    sliced before the concat (the slice copy happens inside the consuming
+    conv's contiguous() — the same cost the cat always paid)."""
+    p = getattr(t, "_pvg_pad", 0)
+    return t[:, :, p:-p, p:-p] if p else t
+
+
 class Encoder64(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
         nf = 64
-        self.c1 = dcgan_conv(nc, nf)              # nc x64x64 -> nf x32x32
-        self.c2 = dcgan_conv(nf, nf * 2)          # -> 128x16x16
-        self.c3 = dcgan_conv(nf * 2, nf * 4)      # -> 256x8x8
-        self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x4x4
+        self.c1 = dcgan_conv(nc, nf, pad_out=True)       # nc x64x64 -> nf x32x32
+        self.c2 = dcgan_conv(nf, nf * 2, pad_out=True)    # -> 128x16x16
+        self.c3 = dcgan_conv(nf * 2, nf * 4, pad_out=True)  # -> 256x8x8
+        self.c4 = dcgan_conv(nf * 4, nf * 8)              # -> 512x4x4 (dense: c5 GEMM)
         self.c5 = FusedSequential(                  # -> dim x1x1
             Conv2d(nf * 8, dim, 4, 1, 0),
             BatchNorm2d(dim),
@@ -62,10 +71,10 @@ class Decoder64(nn.Module):
     def forward(self, inp):
         vec, skip = inp
         d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
-        d2 = self.upc2(torch.cat([d1, skip[3]], 1))
-        d3 = self.upc3(torch.cat([d2, skip[2]], 1))
-        d4 = self.upc4(torch.cat([d3, skip[1]], 1))
-        return self.upc5(torch.cat([d4, skip[0]], 1))
+        d2 = self.upc2(torch.cat([d1, _interior(skip[3])], 1))
+        d3 = self.upc3(torch.cat([d2, _interior(skip[2])], 1))
+        d4 = self.upc4(torch.cat([d3, _interior(skip[1])], 1))
+        return self.upc5(torch.cat([d4, _interior(skip[0])], 1))
 
 
 class Encoder128(nn.Module):
@@ -75,11 +84,11 @@ class Encoder128(nn.Module):
         super().__init__()
         self.dim = dim
         nf = 64
-        self.c1 = dcgan_conv(nc, nf)              # nc x128 -> 64x64
-        self.c2 = dcgan_conv(nf, nf * 2)          # -> 128x32
-        self.c3 = dcgan_conv(nf * 2, nf * 4)      # -> 256x16
-        self.c4 = dcgan_conv(nf * 4, nf * 8)      # -> 512x8
-        self.c5 = dcgan_conv(nf * 8, nf * 8)      # -> 512x4
+        self.c1 = dcgan_conv(nc, nf, pad_out=True)        # nc x128 -> 64x64
+        self.c2 = dcgan_conv(nf, nf * 2, pad_out=True)     # -> 128x32
+        self.c3 = dcgan_conv(nf * 2, nf * 4, pad_out=True)  # -> 256x16
+        self.c4 = dcgan_conv(nf * 4, nf * 8, pad_out=True)  # -> 512x8
+        self.c5 = dcgan_conv(nf * 8, nf * 8)               # -> 512x4 (dense: c6 GEMM)
         self.c6 = FusedSequential(
             Conv2d(nf * 8, dim, 4, 1, 0),
             BatchNorm2d(dim),
@@ -118,8 +127,8 @@ class Decoder128(nn.Module):
     def forward(self, inp):
         vec, skip = inp
         d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
-        d2 = self.upc2(torch.cat([d1, skip[4]], 1))
-        d3 = self.upc3(torch.cat([d2, skip[3]], 1))
-        d4 = self.upc4(torch.cat([d3, skip[2]], 1))
-        d5 = self.upc5(torch.cat([d4, skip[1]], 1))
-        return self.upc6(torch.cat([d5, skip[0]], 1))
+        d2 = self.upc2(torch.cat([d1, _interior(skip[4])], 1))
+        d3 = self.upc3(torch.cat([d2, _interior(skip[3])], 1))
+        d4 = self.upc4(torch.cat([d3, _interior(skip[2])], 1))
+        d5 = self.upc5(torch.cat([d4, _interior(skip[1])], 1))
+        return self.upc6(torch.cat([d5, _interior(skip[0])], 1))

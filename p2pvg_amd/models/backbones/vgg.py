@@ -18,17 +18,29 @@ from ...ops.pool import MaxPool2d, UpsamplingNearest2d
 from .blocks import vgg_layer
 
 
+def _cat_skip(a, b):
+    """Channel-concat whose padded-map marker survives (torch.cat drops
+    python attrs); both operands share the same zero-ring width by
+    construction (upsample out and the encoder skip)."""
+    r = torch.cat([a, b], 1)
+    pa = getattr(a, "_pvg_pad", 0)
+    if pa and pa == getattr(b, "_pvg_pad", 0):
+        r._pvg_pad = pa
+    return r
+
+
 class Encoder64(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
-        self.c1 = FusedSequential(vgg_layer(nc, 64), vgg_layer(64, 64))
-        self.c2 = FusedSequential(vgg_layer(64, 128), vgg_layer(128, 128))
+        P = dict(pad_out=True)
+        self.c1 = FusedSequential(vgg_layer(nc, 64, **P), vgg_layer(64, 64, **P))
+        self.c2 = FusedSequential(vgg_layer(64, 128, **P), vgg_layer(128, 128, **P))
         self.c3 = FusedSequential(
-            vgg_layer(128, 256), vgg_layer(256, 256), vgg_layer(256, 256)
+            vgg_layer(128, 256, **P), vgg_layer(256, 256, **P), vgg_layer(256, 256, **P)
         )
         self.c4 = FusedSequential(
-            vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+            vgg_layer(256, 512, **P), vgg_layer(512, 512, **P), vgg_layer(512, 512, **P)
         )
         self.c5 = FusedSequential(
             Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
@@ -37,9 +49,9 @@ class Encoder64(nn.Module):
 
     def forward(self, x):
         h1 = self.c1(x)
-        h2 = self.c2(self.mp(h1))
-        h3 = self.c3(self.mp(h2))
-        h4 = self.c4(self.mp(h3))
+        h2 = self.c2(self.mp(h1, pad_out=True))
+        h3 = self.c3(self.mp(h2, pad_out=True))
+        h4 = self.c4(self.mp(h3, pad_out=True))
         h5 = self.c5(self.mp(h4))
         return h5.view(-1, self.dim), [h1, h2, h3, h4]
 
@@ -53,15 +65,16 @@ class Decoder64(nn.Module):
             BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
+        P = dict(pad_out=True)
         self.upc2 = FusedSequential(
-            vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 256)
+            vgg_layer(512 * 2, 512, **P), vgg_layer(512, 512, **P), vgg_layer(512, 256, **P)
         )
         self.upc3 = FusedSequential(
-            vgg_layer(256 * 2, 256), vgg_layer(256, 256), vgg_layer(256, 128)
+            vgg_layer(256 * 2, 256, **P), vgg_layer(256, 256, **P), vgg_layer(256, 128, **P)
         )
-        self.upc4 = FusedSequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
+        self.upc4 = FusedSequential(vgg_layer(128 * 2, 128, **P), vgg_layer(128, 64, **P))
         self.upc5 = FusedSequential(
-            vgg_layer(64 * 2, 64),
+            vgg_layer(64 * 2, 64, **P),
             ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
         )
@@ -70,26 +83,27 @@ class Decoder64(nn.Module):
     def forward(self, inp):
         vec, skip = inp
         d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
-        d2 = self.upc2(torch.cat([self.up(d1), skip[3]], 1))
-        d3 = self.upc3(torch.cat([self.up(d2), skip[2]], 1))
-        d4 = self.upc4(torch.cat([self.up(d3), skip[1]], 1))
-        return self.upc5(torch.cat([self.up(d4), skip[0]], 1))
+        d2 = self.upc2(_cat_skip(self.up(d1, pad_out=True), skip[3]))
+        d3 = self.upc3(_cat_skip(self.up(d2, pad_out=True), skip[2]))
+        d4 = self.upc4(_cat_skip(self.up(d3, pad_out=True), skip[1]))
+        return self.upc5(_cat_skip(self.up(d4, pad_out=True), skip[0]))
 
 
 class Encoder128(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
         self.dim = dim
-        self.c1 = FusedSequential(vgg_layer(nc, 64), vgg_layer(64, 64))
-        self.c2 = FusedSequential(vgg_layer(64, 128), vgg_layer(128, 128))
+        P = dict(pad_out=True)
+        self.c1 = FusedSequential(vgg_layer(nc, 64, **P), vgg_layer(64, 64, **P))
+        self.c2 = FusedSequential(vgg_layer(64, 128, **P), vgg_layer(128, 128, **P))
         self.c3 = FusedSequential(
-            vgg_layer(128, 256), vgg_layer(256, 256), vgg_layer(256, 256)
+            vgg_layer(128, 256, **P), vgg_layer(256, 256, **P), vgg_layer(256, 256, **P)
         )
         self.c4 = FusedSequential(
-            vgg_layer(256, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+            vgg_layer(256, 512, **P), vgg_layer(512, 512, **P), vgg_layer(512, 512, **P)
         )
         self.c5 = FusedSequential(
-            vgg_layer(512, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+            vgg_layer(512, 512, **P), vgg_layer(512, 512, **P), vgg_layer(512, 512, **P)
         )
         self.c6 = FusedSequential(
             Conv2d(512, dim, 4, 1, 0), BatchNorm2d(dim), nn.Tanh()
@@ -98,10 +112,10 @@ class Encoder128(nn.Module):
 
     def forward(self, x):
         h1 = self.c1(x)
-        h2 = self.c2(self.mp(h1))
-        h3 = self.c3(self.mp(h2))
-        h4 = self.c4(self.mp(h3))
-        h5 = self.c5(self.mp(h4))
+        h2 = self.c2(self.mp(h1, pad_out=True))
+        h3 = self.c3(self.mp(h2, pad_out=True))
+        h4 = self.c4(self.mp(h3, pad_out=True))
+        h5 = self.c5(self.mp(h4, pad_out=True))
         h6 = self.c6(self.mp(h5))
         return h6.view(-1, self.dim), [h1, h2, h3, h4, h5]
 
@@ -115,18 +129,19 @@ class Decoder128(nn.Module):
             BatchNorm2d(512),
             nn.LeakyReLU(0.2, inplace=True),
         )
+        P = dict(pad_out=True)
         self.upc2 = FusedSequential(
-            vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 512)
+            vgg_layer(512 * 2, 512, **P), vgg_layer(512, 512, **P), vgg_layer(512, 512, **P)
         )
         self.upc3 = FusedSequential(
-            vgg_layer(512 * 2, 512), vgg_layer(512, 512), vgg_layer(512, 256)
+            vgg_layer(512 * 2, 512, **P), vgg_layer(512, 512, **P), vgg_layer(512, 256, **P)
         )
         self.upc4 = FusedSequential(
-            vgg_layer(256 * 2, 256), vgg_layer(256, 256), vgg_layer(256, 128)
+            vgg_layer(256 * 2, 256, **P), vgg_layer(256, 256, **P), vgg_layer(256, 128, **P)
         )
-        self.upc5 = FusedSequential(vgg_layer(128 * 2, 128), vgg_layer(128, 64))
+        self.upc5 = FusedSequential(vgg_layer(128 * 2, 128, **P), vgg_layer(128, 64, **P))
         self.upc6 = FusedSequential(
-            vgg_layer(64 * 2, 64),
+            vgg_layer(64 * 2, 64, **P),
             ConvTranspose2d(64, nc, 3, 1, 1),
             nn.Sigmoid(),
         )
@@ -135,8 +150,8 @@ class Decoder128(nn.Module):
     def forward(self, inp):
         vec, skip = inp
         d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
-        d2 = self.upc2(torch.cat([self.up(d1), skip[4]], 1))
-        d3 = self.upc3(torch.cat([self.up(d2), skip[3]], 1))
-        d4 = self.upc4(torch.cat([self.up(d3), skip[2]], 1))
-        d5 = self.upc5(torch.cat([self.up(d4), skip[1]], 1))
-        return self.upc6(torch.cat([self.up(d5), skip[0]], 1))
+        d2 = self.upc2(_cat_skip(self.up(d1, pad_out=True), skip[4]))
+        d3 = self.upc3(_cat_skip(self.up(d2, pad_out=True), skip[3]))
+        d4 = self.upc4(_cat_skip(self.up(d3, pad_out=True), skip[2]))
+        d5 = self.upc5(_cat_skip(self.up(d4, pad_out=True), skip[1]))
+        return self.upc6(_cat_skip(self.up(d5, pad_out=True), skip[0]))

@@ -196,18 +196,26 @@ class Conv2dNHWCFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int = 0,
-                want_stats: bool = False, w_fwd=None, w_bwd=None):
+                want_stats: bool = False, w_fwd=None, w_bwd=None,
+                in_ring: int = 0, out_ring: int = 0):
         # w receives the weight gradient. With w_fwd/w_bwd (per-step bf16
         # shadows, see refresh_conv_shadows), w is the raw fp32 param and
         # the per-use cast / flip / transpose kernels disappear; without
         # them w must already be bf16 channels_last (legacy direct calls).
+        # in_ring: x is physically (H+2r, W+2r) with a ZERO ring realizing
+        # the conv padding (must equal pad) — gathers become in-bounds and
+        # the glds pipeline applies. out_ring: write the logical output as
+        # the interior of a (HO+2r, WO+2r) map (ring left garbage; the
+        # following bn_act writes the zeros).
         ext = _ext()
         wf = w_fwd if w_fwd is not None else w
         k = wf.shape[2]
         b32 = b.float() if b is not None else None
         stats = None
+        assert in_ring in (0, pad), "in_ring must equal the conv padding"
         # degenerate whole-image conv (k == H, pad 0): plain GEMM
-        gemm = pad == 0 and k == x.shape[2] and k == x.shape[3]
+        gemm = pad == 0 and in_ring == 0 and out_ring == 0 \
+            and k == x.shape[2] and k == x.shape[3]
         if gemm:
             out = torch.mm(_nhwc_flat(x), _nhwc_flat(wf).t())
             if b is not None:
@@ -221,8 +229,15 @@ class Conv2dNHWCFn(torch.autograd.Function):
             ctx.save_for_backward(x, wf, out if act != 0 else None)
             ctx.shadow_bwd = False
         else:
-            out, stats = ext.conv2d_nhwc_fwd(x, wf, b32, stride, pad, act,
-                                             want_stats)
+            oh = ow = oy = 0
+            if out_ring:
+                H = x.shape[2] - 2 * in_ring
+                ho = (H + 2 * pad - k) // stride + 1
+                oh = ow = ho + 2 * out_ring
+                oy = out_ring
+            out, stats = ext.conv2d_nhwc_fwd(
+                x, wf, b32, stride, pad - in_ring, act, want_stats,
+                oh, ow, oy, oy)
             wb = w_bwd if w_bwd is not None else wf
             ctx.save_for_backward(x, wb, out if act != 0 else None)
             ctx.shadow_bwd = w_bwd is not None
@@ -231,6 +246,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
         ctx.act = act
         ctx.wdtype = w.dtype
         ctx.wref, ctx.bref = w, b
+        ctx.rings = (in_ring, out_ring)
         if stats is None or not want_stats:
             stats = torch.empty(0, device=x.device)
         ctx.mark_non_differentiable(stats)
@@ -264,21 +280,28 @@ class Conv2dNHWCFn(torch.autograd.Function):
                     dw = dw.to(ctx.wdtype)
             if ctx.has_bias and ctx.needs_input_grad[2] and wge:
                 db = g2.sum(0, dtype=torch.float32)
-            base = (dx, dw, db, None, None, None, None)
-            return base + (None, None) if ctx.with_shadows else base
+            return (dx, dw, db) + (None,) * 8
 
+        in_ring, out_ring = ctx.rings
         if ctx.needs_input_grad[0]:
+            # dx shape must equal x's (padded) shape; the ring is garbage
+            # the interior-reading consumers (bn backward) ignore
+            oh = x.shape[2] if in_ring else 0
+            ow = x.shape[3] if in_ring else 0
             if stride == 1:
                 wt = w if ctx.shadow_bwd else \
                     w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)
-                dx = ext.conv2d_nhwc_fwd(gout, wt, None, 1, k - 1 - pad, 0,
-                                         False)[0]
+                # gout's zero ring (out_ring) realizes that much of the
+                # dgrad padding k-1-pad
+                dx = ext.conv2d_nhwc_fwd(gout, wt, None, 1,
+                                         k - 1 - pad - out_ring, 0, False,
+                                         oh, ow, in_ring, in_ring)[0]
             else:
                 wt = w if ctx.shadow_bwd else \
                     w.transpose(0, 1).contiguous(memory_format=CL)  # (C,K,k,k)
                 dx = ext.conv2d_nhwc_fracstride(
-                    gout, wt, None, stride, pad, x.shape[2], x.shape[3], 0,
-                    False
+                    gout, wt, None, stride, pad, x.shape[2] - 2 * in_ring,
+                    x.shape[3] - 2 * in_ring, 0, False, out_ring, in_ring
                 )[0]
         if ctx.needs_input_grad[1] and wge:
             wg = _acc_target(ctx.wref)
@@ -286,9 +309,11 @@ class Conv2dNHWCFn(torch.autograd.Function):
                     and wg.is_contiguous(memory_format=CL):
                 # accumulate straight into the managed fp32 .grad: autograd's
                 # per-use AccumulateGrad adds never run for conv weights
-                ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0, wg)
+                ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad - in_ring,
+                                      0, wg, out_ring)
             else:
-                ws = ext.conv2d_nhwc_wgrad(gout, x, k, k, stride, pad, 0)
+                ws = ext.conv2d_nhwc_wgrad(gout, x, k, k, stride,
+                                           pad - in_ring, 0, None, out_ring)
                 if ctx.wdtype == torch.float32:
                     # (B,R,S,A) contiguous permuted to (K,C,k,k) IS the
                     # standard channels_last layout: the fp32 workspace is
@@ -299,8 +324,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
                         .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2] and wge:
             db = _channel_sum(gout, _acc_target(ctx.bref))
-        base = (dx, dw, db, None, None, None, None)
-        return base + (None, None) if ctx.with_shadows else base
+        return (dx, dw, db) + (None,) * 8
 
 
 class ConvT2dNHWCFn(torch.autograd.Function):
@@ -308,18 +332,24 @@ class ConvT2dNHWCFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, b, stride: int, pad: int, act: int = 0,
-                want_stats: bool = False, w_fwd=None, w_bwd=None):
+                want_stats: bool = False, w_fwd=None, w_bwd=None,
+                in_ring: int = 0, out_ring: int = 0):
         # With shadows: w is the raw fp32 param (receives the grad), w_fwd is
         # the pre-flipped/transposed (Co,Ci,k,k) bf16 forward form, w_bwd the
         # plain-cast (Ci,Co,k,k) bf16 form (dgrad + gemm operand).
+        # in_ring: x physically padded with a zero ring (stride-1 convT:
+        # must equal k-1-pad, the flipped conv's padding); out_ring: write
+        # the output as the interior of a padded map.
         ext = _ext()
         wp = w_bwd if w_bwd is not None else w  # plain (Ci,Co,k,k) bf16 form
         k = wp.shape[2]
         co = wp.shape[1]
-        n, _, h, wdt = x.shape
+        n, _, hp, wdtp = x.shape
+        h, wdt = hp - 2 * in_ring, wdtp - 2 * in_ring
         b32 = b.float() if b is not None else None
         stats = None
-        gemm = stride == 1 and pad == 0 and h == 1 and wdt == 1
+        gemm = stride == 1 and pad == 0 and in_ring == 0 and out_ring == 0 \
+            and h == 1 and wdt == 1
         if gemm:
             # 1x1 -> kxk: out[n, y, x, co] = sum_ci in[n,ci] w[ci,co,y,x].
             # The channels_last (Ci,Co,k,k) weight is physically (Ci,k,k,Co),
@@ -332,17 +362,23 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                 stats = _gemm_stats(out.reshape(n * k * k, co)).unsqueeze(0)
             out = out.view(n, k, k, co).permute(0, 3, 1, 2)
         elif stride == 1:
+            assert in_ring in (0, k - 1 - pad), "convT in_ring mismatch"
             wt = w_fwd if w_fwd is not None else \
                 w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)  # (Co,Ci,k,k)
-            out, stats = ext.conv2d_nhwc_fwd(x, wt, b32, 1, k - 1 - pad, act,
-                                             want_stats)
+            ho = h + k - 1 - 2 * pad
+            oh = ho + 2 * out_ring if out_ring else 0
+            out, stats = ext.conv2d_nhwc_fwd(x, wt, b32, 1,
+                                             k - 1 - pad - in_ring, act,
+                                             want_stats, oh, oh,
+                                             out_ring, out_ring)
         else:
             oh = (h - 1) * stride - 2 * pad + k
             ow = (wdt - 1) * stride - 2 * pad + k
             wt = w_fwd if w_fwd is not None else \
                 w.transpose(0, 1).contiguous(memory_format=CL)  # (Co, Ci, k, k)
             out, stats = ext.conv2d_nhwc_fracstride(x, wt, b32, stride, pad,
-                                                    oh, ow, act, want_stats)
+                                                    oh, ow, act, want_stats,
+                                                    in_ring, out_ring)
         ctx.save_for_backward(x, wp, out if act != 0 else None)
         ctx.shadow_bwd = w_bwd is not None
         ctx.with_shadows = w_fwd is not None or w_bwd is not None
@@ -350,6 +386,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         ctx.act = act
         ctx.wdtype = w.dtype
         ctx.wref, ctx.bref = w, b
+        ctx.rings = (in_ring, out_ring)
         if stats is None or not want_stats:
             stats = torch.empty(0, device=x.device)
         ctx.mark_non_differentiable(stats)
@@ -383,20 +420,27 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                     dw = dw.to(ctx.wdtype)
             if ctx.has_bias and ctx.needs_input_grad[2] and wge:
                 db = gout.sum(dim=(0, 2, 3), dtype=torch.float32)
-            base = (dx, dw, db, None, None, None, None)
-            return base + (None, None) if ctx.with_shadows else base
+            return (dx, dw, db) + (None,) * 8
 
+        in_ring, out_ring = ctx.rings
         if ctx.needs_input_grad[0]:
-            # dgrad of convT = plain conv with the untransposed weight
+            # dgrad of convT = plain conv with the untransposed weight;
+            # gout's zero ring realizes that much of the conv padding, dx
+            # is written as the interior of x's padded shape (garbage ring)
             wl = w if ctx.shadow_bwd else w.contiguous(memory_format=CL)
-            dx = ext.conv2d_nhwc_fwd(gout, wl, None, stride, pad, 0, False)[0]
+            oh = x.shape[2] if in_ring else 0
+            ow = x.shape[3] if in_ring else 0
+            dx = ext.conv2d_nhwc_fwd(gout, wl, None, stride, pad - out_ring,
+                                     0, False, oh, ow, in_ring, in_ring)[0]
         if ctx.needs_input_grad[1] and wge:
             wg = _acc_target(ctx.wref)
             if wg is not None and ctx.wdtype == torch.float32 \
                     and wg.is_contiguous(memory_format=CL):
-                ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad, 0, wg)
+                ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad - out_ring,
+                                      0, wg, in_ring)
             else:
-                ws = ext.conv2d_nhwc_wgrad(x, gout, k, k, stride, pad, 0)
+                ws = ext.conv2d_nhwc_wgrad(x, gout, k, k, stride,
+                                           pad - out_ring, 0, None, in_ring)
                 if ctx.wdtype == torch.float32:
                     dw = ws.permute(0, 3, 1, 2)
                 else:
@@ -404,8 +448,15 @@ class ConvT2dNHWCFn(torch.autograd.Function):
                         .contiguous(memory_format=CL)
         if ctx.has_bias and ctx.needs_input_grad[2] and wge:
             db = _channel_sum(gout, _acc_target(ctx.bref))
-        base = (dx, dw, db, None, None, None, None)
-        return base + (None, None) if ctx.with_shadows else base
+        return (dx, dw, db) + (None,) * 8
+
+
+def interior_view(t: torch.Tensor) -> torch.Tensor:
+    """Logical (interior) view of a padded zero-ring map; identity on dense
+    tensors. Safety net for padded tensors reaching consumers that don't
+    speak the ring protocol."""
+    p = getattr(t, "_pvg_pad", 0)
+    return t[:, :, p:-p, p:-p] if p else t
 
 
 def _use_hip_path(x: torch.Tensor) -> bool:
@@ -433,15 +484,19 @@ class Conv2d(nn.Conv2d):
             and self.dilation == (1, 1)
             and self.groups == 1
         ):
+            in_ring = getattr(x, "_pvg_pad", 0)
+            if in_ring and in_ring != self.padding[0]:
+                x = interior_view(x)
+                in_ring = 0
             with torch.autocast("cuda", enabled=False):
                 xl = _to_cl_bf16(x)
                 sh = _conv_shadows(self)
                 out, _ = Conv2dNHWCFn.apply(
                     xl, self.weight, self.bias, self.stride[0],
-                    self.padding[0], 0, False, sh["f"], sh["b"]
+                    self.padding[0], 0, False, sh["f"], sh["b"], in_ring, 0
                 )
                 return out
-        return super().forward(x)
+        return super().forward(interior_view(x))
 
 
 class ConvTranspose2d(nn.ConvTranspose2d):
@@ -460,12 +515,19 @@ class ConvTranspose2d(nn.ConvTranspose2d):
             and self.groups == 1
             and self.output_padding == (0, 0)
         ):
+            k = self.kernel_size[0]
+            want = k - 1 - self.padding[0] if self.stride[0] == 1 else \
+                self.padding[0]
+            in_ring = getattr(x, "_pvg_pad", 0)
+            if in_ring and (self.stride[0] != 1 or in_ring != want):
+                x = interior_view(x)
+                in_ring = 0
             with torch.autocast("cuda", enabled=False):
                 xl = _to_cl_bf16(x)
                 sh = _conv_shadows(self)
                 out, _ = ConvT2dNHWCFn.apply(
                     xl, self.weight, self.bias, self.stride[0],
-                    self.padding[0], 0, False, sh["f"], sh["b"]
+                    self.padding[0], 0, False, sh["f"], sh["b"], in_ring, 0
                 )
                 return out
-        return super().forward(x, output_size)
+        return super().forward(interior_view(x), output_size)

@@ -450,7 +450,8 @@ void check_nhwc_bf16(const torch::Tensor& t, const char* name) {
 std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
                                             c10::optional<torch::Tensor> bias,
                                             long stride, long pad, long act,
-                                            bool want_stats) {
+                                            bool want_stats, long oh, long ow,
+                                            long oy0, long ox0) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
@@ -463,10 +464,13 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
   if (conv2d_glds_eligible(C, K, R, stride, pad) &&
       ((long)Nb * HO * WO / 256) * (K / (K % 128 ? 64 : 128)) >= 200 &&
       glds_enabled()) {
-    return conv2d_glds_fwd(in, w, bias, stride, act, want_stats, 0, 0, 0, 0);
+    return conv2d_glds_fwd(in, w, bias, stride, act, want_stats, oh, ow, oy0,
+                           ox0);
   }
 
-  auto out = torch::empty({Nb, K, HO, WO},
+  const int OH = oh > 0 ? (int)oh : HO;
+  const int OW = ow > 0 ? (int)ow : WO;
+  auto out = torch::empty({Nb, K, OH, OW},
                           in.options().memory_format(at::MemoryFormat::ChannelsLast));
   if (bias.has_value()) {
     CHECK_INPUT(bias.value());
@@ -474,8 +478,9 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
   }
   ConvArgs a{};
   a.Nb = Nb; a.H = H; a.W = W; a.C = C; a.K = K;
-  a.HO = HO; a.WO = WO; a.OH = HO; a.OW = WO;
+  a.HO = HO; a.WO = WO; a.OH = OH; a.OW = OW;
   a.act = (int)act; a.wk = R; a.oys = 1;
+  a.oy0[0] = (int)oy0; a.ox0[0] = (int)ox0;
   a.padh[0] = (int)pad; a.padw[0] = (int)pad;
   a.stats = nullptr;
   torch::Tensor stats_out;
@@ -504,7 +509,7 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
 std::vector<torch::Tensor> conv2d_nhwc_fracstride(
     torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
     long up_stride, long up_pad, long OH, long OW, long act,
-    bool want_stats) {
+    bool want_stats, long in_ring, long out_ring) {
   check_nhwc_bf16(in, "in");
   check_nhwc_bf16(w, "w");
   const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
@@ -513,8 +518,9 @@ std::vector<torch::Tensor> conv2d_nhwc_fracstride(
   TORCH_CHECK(up_stride == 2, "fracstride: stride 2 only");
   const int st = (int)up_stride, pad = (int)up_pad;
 
-  auto out = torch::empty({Nb, K, OH, OW},
-                          in.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto out = torch::empty(
+      {Nb, K, OH + 2 * out_ring, OW + 2 * out_ring},
+      in.options().memory_format(at::MemoryFormat::ChannelsLast));
   if (bias.has_value()) {
     CHECK_INPUT(bias.value());
     TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be fp32");
@@ -522,7 +528,7 @@ std::vector<torch::Tensor> conv2d_nhwc_fracstride(
 
   ConvArgs a{};
   a.Nb = Nb; a.H = H; a.W = W; a.C = C; a.K = K;
-  a.OH = (int)OH; a.OW = (int)OW;
+  a.OH = (int)(OH + 2 * out_ring); a.OW = (int)(OW + 2 * out_ring);
   a.act = (int)act; a.wk = WK; a.oys = st;
   a.stats = nullptr;
   torch::Tensor stats_out;
@@ -556,16 +562,17 @@ std::vector<torch::Tensor> conv2d_nhwc_fracstride(
     for (int px = 0; px < st; ++px) {
       const int pi = p * st + px;
       for (int i = 0; i < nt; ++i) a.rmap[pi][i] = taps[i][0];
-      a.padh[pi] = ipad;
-      a.oy0[pi] = p;
+      a.padh[pi] = ipad - (int)in_ring;   // padded input: shift gathers +ring
+      a.oy0[pi] = p + (int)out_ring;      // padded output: interior origin
     }
     for (int py = 0; py < st; ++py) {
       const int pi = py * st + p;
       for (int i = 0; i < nt; ++i) a.smap[pi][i] = taps[i][0];
-      a.padw[pi] = ipad;
-      a.ox0[pi] = p;
+      a.padw[pi] = ipad - (int)in_ring;
+      a.ox0[pi] = p + (int)out_ring;
     }
     // compact grid dims are the same for all parities when OH even
+    // (OH here is the LOGICAL output height; rings only move origins)
     const int hoc = ((int)OH - 1 - p) / st + 1;
     if (HOc < 0) HOc = hoc; else TORCH_CHECK(hoc == HOc, "uneven parity grid");
     const int woc = ((int)OW - 1 - p) / st + 1;

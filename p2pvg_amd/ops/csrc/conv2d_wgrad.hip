@@ -57,13 +57,16 @@ __device__ __forceinline__ bf16x8 tr16_combine(u16x4 lo, u16x4 hi) {
 
 // BTB/BTA: channel tiles per side (64 or 128, independent); wave tile
 // (BTB/2 x BTA/2), (BTB/32 x BTA/32) fragments per wave.
+// yring > 0: Y is physically (N, HO+2r, WO+2r, B) with the logical content
+// in the interior — pixel indices stay over the logical HOxWO grid so no
+// FLOPs are spent on the ring.
 template <int BTB, int BTA>
 __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
-    const __bf16* __restrict__ Y,  // (N, HO, WO, B)
+    const __bf16* __restrict__ Y,  // (N, HO, WO, B) (+ ring, see above)
     const __bf16* __restrict__ X,  // (N, H, W, A)
-    float* __restrict__ ws,        // (B, R, S, A) fp32, pre-zeroed
+    float* __restrict__ ws,        // (sp, B, R, S, A) fp32 slab stores
     int Nb, int HO, int WO, int B, int H, int W, int A, int R, int S,
-    int STRIDE, int PAD, int p_per_slab) {
+    int STRIDE, int PAD, int p_per_slab, int yring) {
   __shared__ __align__(16) char lds[PCH * (BTB + BTA) * 2];
   char* yt = lds;                        // [PCH/4][BTB/16][4][16] subtiles
   char* xt = lds + PCH * BTB * 2;        // [PCH/4][BTA/16][4][16] subtiles
@@ -98,6 +101,8 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
   // T14 pipeline: issue chunk t+1's global loads, MFMA chunk t from LDS,
   // write t+1 after the barrier. Pixel meta is computed inline per staging
   // slot (few int divides, hidden under the loads).
+  const int HOp = HO + 2 * yring;
+  const int WOp = WO + 2 * yring;
   auto load_chunk = [&](int p0) {
 #pragma unroll
     for (int it = 0; it < SLY; ++it) {
@@ -107,7 +112,15 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_kernel(
       const int pix = p0 + pix_l;
       bf16x8 vy = {};
       if (pix < p_end && b0 + ch0 < B) {
-        const __bf16* src = Y + (long)pix * B + b0 + ch0;
+        long yoff = (long)pix * B;
+        if (yring > 0) {
+          const int n = pix / (HO * WO);
+          const int rem = pix - n * (HO * WO);
+          const int ho = rem / WO;
+          const int wo = rem - ho * WO;
+          yoff = (((long)n * HOp + ho + yring) * WOp + wo + yring) * B;
+        }
+        const __bf16* src = Y + yoff + b0 + ch0;
         if (b0 + ch0 + 8 <= B) {
           vy = *reinterpret_cast<const bf16x8*>(src);
         } else {
@@ -254,14 +267,15 @@ __global__ __launch_bounds__(256) void wgrad_combine_kernel(
 // accumulation adds entirely (the ~29 uses per weight per step land here).
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp,
-                                c10::optional<torch::Tensor> acc) {
+                                c10::optional<torch::Tensor> acc, long yring) {
   TORCH_CHECK(Y.is_cuda() && Y.scalar_type() == torch::kBFloat16 &&
                   Y.is_contiguous(at::MemoryFormat::ChannelsLast),
               "Y must be bf16 channels_last GPU");
   TORCH_CHECK(X.is_cuda() && X.scalar_type() == torch::kBFloat16 &&
                   X.is_contiguous(at::MemoryFormat::ChannelsLast),
               "X must be bf16 channels_last GPU");
-  const int Nb = Y.size(0), B = Y.size(1), HO = Y.size(2), WO = Y.size(3);
+  const int Nb = Y.size(0), B = Y.size(1);
+  const int HO = Y.size(2) - 2 * (int)yring, WO = Y.size(3) - 2 * (int)yring;
   const int A = X.size(1), H = X.size(2), W = X.size(3);
   TORCH_CHECK(X.size(0) == Nb, "batch mismatch");
   const long E = (long)B * R * S * A;
@@ -290,7 +304,7 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                      stream, reinterpret_cast<const __bf16*>(Y.data_ptr()),    \
                      reinterpret_cast<const __bf16*>(X.data_ptr()),            \
                      ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,     \
-                     (int)S, (int)stride, (int)pad, p_per_slab)
+                     (int)S, (int)stride, (int)pad, p_per_slab, (int)yring)
   if (BTB == 128 && BTA == 128) WGRAD_LAUNCH(128, 128);
   else if (BTB == 128) WGRAD_LAUNCH(128, 64);
   else if (BTA == 128) WGRAD_LAUNCH(64, 128);

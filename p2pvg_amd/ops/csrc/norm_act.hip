@@ -114,13 +114,17 @@ __global__ void bn_eval_prep_kernel(const float* __restrict__ running_mean,
   shift_out[c] = beta[c] - m * gamma[c] * inv;
 }
 
-// y = act(x*scale[c] + shift[c]); vectors of 8 bf16 along C
+// y = act(x*scale[c] + shift[c]); vectors of 8 bf16 along C.
 // C/8 divides BLOCK and the grid stride, so each thread's channel group is
 // fixed: per-channel scalars are preloaded into registers once.
+// ring > 0: x and y are (N, C, H+2r, W+2r) padded maps whose logical content
+// is the interior — the walk covers the FULL padded tensor, writing ZERO at
+// ring positions (the next conv gathers the ring as its padding) and
+// ignoring x's ring (which holds garbage from interior-only conv writes).
 __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
     const __bf16* __restrict__ x, const float* __restrict__ scale,
     const float* __restrict__ shift, __bf16* __restrict__ y, long nvec, int C,
-    int act) {
+    int act, int Hp, int Wp, int ring) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float sc[8], sh[8];
@@ -131,6 +135,16 @@ __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
   }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
+    if (ring > 0) {
+      const long pix = i / cvec;
+      const long rem = pix % ((long)Hp * Wp);
+      const int yy = (int)(rem / Wp);
+      const int xx = (int)(rem - (long)yy * Wp);
+      if (yy < ring || yy >= Hp - ring || xx < ring || xx >= Wp - ring) {
+        *reinterpret_cast<bf16x8*>(y + i * 8) = bf16x8{};
+        continue;
+      }
+    }
     bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 o;
 #pragma unroll
@@ -151,7 +165,7 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ rows,  // (gridDim.x, 2, C) per-block partials (stores)
-    long nvec, int C, int act) {
+    long nvec, int C, int act, int Hp, int Wp, int ring) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float p1[8], p2[8];
@@ -168,6 +182,14 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
 
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
+    if (ring > 0) {
+      const long pix = i / cvec;
+      const long rem = pix % ((long)Hp * Wp);
+      const int yy = (int)(rem / Wp);
+      const int xx = (int)(rem - (long)yy * Wp);
+      if (yy < ring || yy >= Hp - ring || xx < ring || xx >= Wp - ring)
+        continue;   // x ring is garbage, dy ring grads are for constants
+    }
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
 #pragma unroll
@@ -333,7 +355,8 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ scale, const float* __restrict__ red,
-    __bf16* __restrict__ dx, long nvec, int C, int act, float inv_count) {
+    __bf16* __restrict__ dx, long nvec, int C, int act, float inv_count,
+    int Hp, int Wp, int ring) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float mn[8], is[8], ga[8], be[8], sc[8], r1[8], r2[8];
@@ -349,6 +372,17 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_kernel(
   }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
+    if (ring > 0) {
+      const long pix = i / cvec;
+      const long rem = pix % ((long)Hp * Wp);
+      const int yy = (int)(rem / Wp);
+      const int xx = (int)(rem - (long)yy * Wp);
+      if (yy < ring || yy >= Hp - ring || xx < ring || xx >= Wp - ring) {
+        // the previous conv's dgrad GATHERS dx's ring: must be zero
+        *reinterpret_cast<bf16x8*>(dx + i * 8) = bf16x8{};
+        continue;
+      }
+    }
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
     bf16x8 o;
@@ -410,12 +444,14 @@ std::vector<torch::Tensor> bn_act_fwd_train(
     torch::Tensor x, torch::Tensor stats, torch::Tensor gamma,
     torch::Tensor beta, c10::optional<torch::Tensor> running_mean,
     c10::optional<torch::Tensor> running_var, double momentum, double eps,
-    long act) {
+    long act, long ring) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
               x.is_contiguous(at::MemoryFormat::ChannelsLast));
   const int C = x.size(1);
   TORCH_CHECK(C % 8 == 0, "bn_act: C must be a multiple of 8");
-  const long count = x.numel() / C;
+  const int Hp = x.size(2), Wp = x.size(3);
+  const long count =
+      (long)x.size(0) * (Hp - 2 * ring) * (Wp - 2 * ring);
   auto f32 = x.options().dtype(torch::kFloat32);
   auto mean = torch::empty({C}, f32);
   auto invstd = torch::empty({C}, f32);
@@ -442,17 +478,20 @@ std::vector<torch::Tensor> bn_act_fwd_train(
   hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK), 0,
                      stream, reinterpret_cast<const __bf16*>(x.data_ptr()),
                      scale.data_ptr<float>(), shift.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C, (int)act);
+                     reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C, (int)act,
+                     Hp, Wp, (int)ring);
   return {y, mean, invstd, scale};
 }
 
 torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
                               torch::Tensor beta, torch::Tensor running_mean,
-                              torch::Tensor running_var, double eps, long act) {
+                              torch::Tensor running_var, double eps, long act,
+                              long ring) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16 &&
               x.is_contiguous(at::MemoryFormat::ChannelsLast));
   const int C = x.size(1);
   TORCH_CHECK(C % 8 == 0, "bn_act: C must be a multiple of 8");
+  const int Hp = x.size(2), Wp = x.size(3);
   auto f32 = x.options().dtype(torch::kFloat32);
   auto mean = torch::empty({C}, f32);
   auto invstd = torch::empty({C}, f32);
@@ -470,7 +509,8 @@ torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
   hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK), 0,
                      stream, reinterpret_cast<const __bf16*>(x.data_ptr()),
                      scale.data_ptr<float>(), shift.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C, (int)act);
+                     reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C, (int)act,
+                     Hp, Wp, (int)ring);
   return y;
 }
 
@@ -484,9 +524,12 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                                       torch::Tensor gamma, torch::Tensor beta,
                                       torch::Tensor scale, long act,
                                       c10::optional<torch::Tensor> dgamma_acc,
-                                      c10::optional<torch::Tensor> dbeta_acc) {
+                                      c10::optional<torch::Tensor> dbeta_acc,
+                                      long ring) {
   const int C = x.size(1);
-  const long count = x.numel() / C;
+  const int Hp = x.size(2), Wp = x.size(3);
+  const long count =
+      (long)x.size(0) * (Hp - 2 * ring) * (Wp - 2 * ring);
   auto f32 = x.options().dtype(torch::kFloat32);
   auto dx = torch::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -500,7 +543,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                      reinterpret_cast<const __bf16*>(dy.data_ptr()),
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                     rows.data_ptr<float>(), nvec, C, (int)act);
+                     rows.data_ptr<float>(), nvec, C, (int)act, Hp, Wp,
+                     (int)ring);
   const int accumulate = dgamma_acc.has_value() ? 1 : 0;
   torch::Tensor dgamma =
       accumulate ? dgamma_acc.value() : torch::empty({C}, f32);
@@ -523,7 +567,7 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                      gamma.data_ptr<float>(), beta.data_ptr<float>(),
                      scale.data_ptr<float>(), red.data_ptr<float>(),
                      reinterpret_cast<__bf16*>(dx.data_ptr()), nvec, C,
-                     (int)act, (float)(1.0 / count));
+                     (int)act, (float)(1.0 / count), Hp, Wp, (int)ring);
   return {dx, dgamma, dbeta};
 }
 

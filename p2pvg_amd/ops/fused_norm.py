@@ -42,16 +42,19 @@ class FusedBNActFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, stats, gamma, beta, running_mean, running_var,
-                momentum: float, eps: float, act: int):
+                momentum: float, eps: float, act: int, ring: int = 0):
+        # ring > 0: x/y are padded maps — the kernels walk the interior and
+        # write y's ring as ZEROS (it becomes the next conv's padding)
         ext = _ext()
         gamma32 = gamma.float()
         beta32 = beta.float()
         y, mean, invstd, scale = ext.bn_act_fwd_train(
             x, stats, gamma32, beta32, running_mean, running_var,
-            momentum, eps, act,
+            momentum, eps, act, ring,
         )
         ctx.save_for_backward(x, mean, invstd, gamma32, beta32, scale)
         ctx.act = act
+        ctx.ring = ring
         ctx.gref, ctx.bref = gamma, beta
         return y
 
@@ -64,32 +67,41 @@ class FusedBNActFn(torch.autograd.Function):
         dy = dy.contiguous(memory_format=CL)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
+        r = ctx.ring
         if not weight_grads_enabled():
             # phase-2 traversal: dx only, param grads are discarded anyway
             dx, _, _ = ext.bn_act_bwd(
-                x, dy, mean, invstd, gamma32, beta32, scale, ctx.act
+                x, dy, mean, invstd, gamma32, beta32, scale, ctx.act,
+                None, None, r,
             )
-            return dx, None, None, None, None, None, None, None, None
+            return dx, None, None, None, None, None, None, None, None, None
         ga = _acc_target(ctx.gref)
         ba = _acc_target(ctx.bref)
         if ga is not None and ba is not None:
             # accumulate dgamma/dbeta straight into the managed .grad buffers
             dx, _, _ = ext.bn_act_bwd(
-                x, dy, mean, invstd, gamma32, beta32, scale, ctx.act, ga, ba
+                x, dy, mean, invstd, gamma32, beta32, scale, ctx.act, ga, ba, r
             )
-            return dx, None, None, None, None, None, None, None, None
+            return dx, None, None, None, None, None, None, None, None, None
         dx, dgamma, dbeta = ext.bn_act_bwd(
-            x, dy, mean, invstd, gamma32, beta32, scale, ctx.act
+            x, dy, mean, invstd, gamma32, beta32, scale, ctx.act, None, None, r
         )
-        return dx, None, dgamma, dbeta, None, None, None, None, None
+        return dx, None, dgamma, dbeta, None, None, None, None, None, None
 
 
-def fused_conv_bn_act(x, conv, bn, act: int):
-    """conv (with stats epilogue) -> fused BN+act. Training and eval modes."""
+def fused_conv_bn_act(x, conv, bn, act: int, pad_out: bool = False):
+    """conv (with stats epilogue) -> fused BN+act. Training and eval modes.
+
+    pad_out: emit y as a PADDED map whose zero ring realizes the NEXT conv's
+    padding (its gathers become in-bounds -> glds staging pipeline). An
+    incoming padded x (attribute `_pvg_pad` set by the previous producer)
+    routes the conv itself through the in-bounds path."""
     from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _conv_shadows, _to_cl_bf16
 
     ext = _ext()
     training = bn.training
+    in_ring = getattr(x, "_pvg_pad", 0)
+    out_ring = 1 if pad_out else 0
     with torch.autocast("cuda", enabled=False):
         xl = _to_cl_bf16(x)
         sh = _conv_shadows(conv)
@@ -101,12 +113,12 @@ def fused_conv_bn_act(x, conv, bn, act: int):
         if isinstance(conv, nn.ConvTranspose2d):
             out, stats = ConvT2dNHWCFn.apply(
                 xl, conv.weight, None, conv.stride[0], conv.padding[0],
-                0, want_stats, sh["f"], sh["b"]
+                0, want_stats, sh["f"], sh["b"], in_ring, out_ring
             )
         else:
             out, stats = Conv2dNHWCFn.apply(
                 xl, conv.weight, None, conv.stride[0], conv.padding[0],
-                0, want_stats, sh["f"], sh["b"]
+                0, want_stats, sh["f"], sh["b"], in_ring, out_ring
             )
         if not want_stats:
             stats = None
@@ -116,35 +128,41 @@ def fused_conv_bn_act(x, conv, bn, act: int):
                 # momentum=None and checkpoint parity); device-side add_ stays
                 # hipGraph-capturable
                 bn.num_batches_tracked.add_(1)
-            return FusedBNActFn.apply(
+            y = FusedBNActFn.apply(
                 out, stats, bn.weight, bn.bias,
                 bn.running_mean if bn.track_running_stats else None,
                 bn.running_var if bn.track_running_stats else None,
                 bn.momentum if bn.momentum is not None else 0.1,
-                bn.eps, act,
+                bn.eps, act, out_ring,
             )
-        return ext.bn_act_fwd_eval(
-            out, bn.weight.float(), bn.bias.float(), bn.running_mean,
-            bn.running_var, bn.eps, act,
-        )
+        else:
+            y = ext.bn_act_fwd_eval(
+                out, bn.weight.float(), bn.bias.float(), bn.running_mean,
+                bn.running_var, bn.eps, act, out_ring,
+            )
+        if out_ring:
+            y._pvg_pad = out_ring
+        return y
 
 
 def fused_conv_act(x, conv, act: int):
-    """conv with the activation fused straight into the epilogue (no BN)."""
+    """conv with the activation fused straight into the epilogue (no BN).
+    Output is always dense; a padded input (attr) is consumed in-bounds."""
     from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _conv_shadows, _to_cl_bf16
 
+    in_ring = getattr(x, "_pvg_pad", 0)
     with torch.autocast("cuda", enabled=False):
         xl = _to_cl_bf16(x)
         sh = _conv_shadows(conv)
         if isinstance(conv, nn.ConvTranspose2d):
             out, _ = ConvT2dNHWCFn.apply(
                 xl, conv.weight, conv.bias, conv.stride[0], conv.padding[0],
-                act, False, sh["f"], sh["b"]
+                act, False, sh["f"], sh["b"], in_ring, 0
             )
         else:
             out, _ = Conv2dNHWCFn.apply(
                 xl, conv.weight, conv.bias, conv.stride[0], conv.padding[0],
-                act, False, sh["f"], sh["b"]
+                act, False, sh["f"], sh["b"], in_ring, 0
             )
         return out
 
@@ -166,10 +184,31 @@ def _conv_supported(conv, x) -> bool:
     )
 
 
+def _consumes_ring(m) -> bool:
+    """Does module m's conv accept a padded (zero-ring) input in place of its
+    own padding? True for the supported pad-1 geometries (k3s1p1, k4s2p1,
+    and convT k3s1p1 whose flipped-conv pad is 1)."""
+    if isinstance(m, nn.ConvTranspose2d):
+        return (m.kernel_size[0] == 3 and m.stride[0] == 1
+                and m.padding[0] == 1)
+    if isinstance(m, nn.Conv2d):
+        return m.padding[0] == 1 and m.kernel_size[0] in (3, 4)
+    return False
+
+
 class FusedSequential(nn.Sequential):
     """nn.Sequential that fuses (conv [, batchnorm] [, activation]) chains on
     the gfx950 path. Same children indices -> same state_dict keys. Greedy
-    scan, so trailing conv+act pairs inside longer chains fuse too."""
+    scan, so trailing conv+act pairs inside longer chains fuse too.
+
+    pad_out (attr `_pvg_pad_out`, set by the backbone): emit the LAST fused
+    block's output as a padded zero-ring map for the consumer conv outside
+    this Sequential. Inside the chain, a block pads out whenever the next
+    module's conv consumes a ring."""
+
+    def __init__(self, *args, pad_out: bool = False):
+        super().__init__(*args)
+        self._pvg_pad_out = pad_out
 
     def forward(self, x):
         from .conv import _use_hip_path
@@ -190,7 +229,12 @@ class FusedSequential(nn.Sequential):
                     and mods[i + 1].num_features % 8 == 0
                     and 256 % (mods[i + 1].num_features // 8) == 0
                 ):
-                    x = fused_conv_bn_act(x, m, mods[i + 1], act_code(mods[i + 2]))
+                    if i + 3 < len(mods):
+                        pad_out = _consumes_ring(mods[i + 3])
+                    else:
+                        pad_out = getattr(self, "_pvg_pad_out", False)
+                    x = fused_conv_bn_act(x, m, mods[i + 1],
+                                          act_code(mods[i + 2]), pad_out)
                     i += 3
                     continue
                 if i + 1 < len(mods) and act_code(mods[i + 1]) is not None:

@@ -56,7 +56,7 @@ def test_conv2d_autograd_matches(N, C, H, W, K, ks, st, pad):
 
     _run_pair(
         lambda xx, ww, bb: torch.nn.functional.conv2d(xx, ww, bb, stride=st, padding=pad),
-        lambda xx, ww, bb: Conv2dNHWCFn.apply(xx, ww, bb, st, pad, 0, False)[0],
+        lambda xx, ww, bb: Conv2dNHWCFn.apply(xx, ww, bb, st, pad, 0, False, None, None, 0, 0)[0],
         x, w, b, tol=0.05,
     )
 
@@ -81,7 +81,7 @@ def test_convtranspose2d_autograd_matches(N, Ci, H, W, Co, ks, st, pad):
     _run_pair(
         lambda xx, ww, bb: torch.nn.functional.conv_transpose2d(
             xx, ww, bb, stride=st, padding=pad),
-        lambda xx, ww, bb: ConvT2dNHWCFn.apply(xx, ww, bb, st, pad, 0, False)[0],
+        lambda xx, ww, bb: ConvT2dNHWCFn.apply(xx, ww, bb, st, pad, 0, False, None, None, 0, 0)[0],
         x, w, b, tol=0.05,
     )
 

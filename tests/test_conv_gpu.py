@@ -4,6 +4,8 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
+CL = torch.channels_last
+
 
 @pytest.fixture(scope="module")
 def ext():
