@@ -44,7 +44,7 @@ class _GraphEntry:
 
 class GraphedTrainStep:
     def __init__(self, model, amp_dtype: Optional[torch.dtype] = None,
-                 warmup_iters: int = 1, max_graphs: int = 16):
+                 warmup_iters: int = 0, max_graphs: int = 16):
         self.model = model
         self.amp_dtype = amp_dtype
         self.warmup_iters = warmup_iters
@@ -152,6 +152,28 @@ class GraphedTrainStep:
                     else:
                         v.zero_()
 
+    def _dry_run(self, entry: _GraphEntry, plan: StepPlan):
+        """One eager step at a TINY batch (2 samples) on a side stream.
+
+        Materializes every lazy one-time initialization (hipBLASLt handles
+        and workspaces, philox state, extension statics) WITHOUT running a
+        full-size eager step first: the full-size warmup's ~20k transient
+        allocations fragmented the caching allocator so badly (observed
+        135 GB reserved-but-unallocated) that the subsequent capture pool
+        OOM'd and the whole bench fell back to eager."""
+        tiny = _GraphEntry()
+        tiny.prev_buf = entry.prev_buf[:, :2].contiguous()
+        tiny.cur_buf = entry.cur_buf[:, :2].contiguous()
+        tiny.tun = entry.tun.clone()
+        tiny.dts = entry.dts.clone()
+        tiny.loss_out = torch.zeros(4, device=entry.loss_out.device)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._inner(tiny, plan)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
     def _capture(self, plan: StepPlan, prev: torch.Tensor, cur: torch.Tensor,
                  tun: torch.Tensor, dts: torch.Tensor) -> _GraphEntry:
         entry = _GraphEntry()
@@ -163,7 +185,11 @@ class GraphedTrainStep:
 
         self._materialize_persistent_state()
         snap = self._snapshot_train_state()
-        # warmup on a side stream (standard CUDAGraph recipe)
+        if not getattr(self, "_lazy_init_done", False):
+            self._dry_run(entry, plan)
+            self._lazy_init_done = True
+        # optional full-size warmup on a side stream (the standard CUDAGraph
+        # recipe; persistent state is already materialized, so default 0)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
