@@ -135,16 +135,6 @@ __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
   }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    if (ring > 0) {
-      const long pix = i / cvec;
-      const long rem = pix % ((long)Hp * Wp);
-      const int yy = (int)(rem / Wp);
-      const int xx = (int)(rem - (long)yy * Wp);
-      if (yy < ring || yy >= Hp - ring || xx < ring || xx >= Wp - ring) {
-        *reinterpret_cast<bf16x8*>(y + i * 8) = bf16x8{};
-        continue;
-      }
-    }
     bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 o;
 #pragma unroll
@@ -153,6 +143,51 @@ __global__ __launch_bounds__(BLOCK) void bn_act_fwd_kernel(
       o[j] = (__bf16)act_fwd(vv, act);
     }
     *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+  }
+}
+
+// ------ ring > 0 row-walk variants ------
+// One padded row (n, yy) per grid-stride block step; within a row the only
+// index math is shifts/masks (cvec = C/8 is a power of two by the C/8 |
+// 256 dispatch gate). Ring positions are written as zeros (fwd y / bwd dx)
+// or skipped (reduce) — no 64-bit divides anywhere in the element loop
+// (the first ring implementation's per-vector i/cvec, %HpWp, /Wp chains
+// were 64-bit software divisions and dominated the padded step).
+
+__global__ __launch_bounds__(BLOCK) void bn_act_fwd_ring_kernel(
+    const __bf16* __restrict__ x, const float* __restrict__ scale,
+    const float* __restrict__ shift, __bf16* __restrict__ y, int NROWS,
+    int Hp, int Wp, int C, int act, int ring, int cvec_sh) {
+  const int cvec = 1 << cvec_sh;
+  const int rowvecs = Wp * C / 8;
+  const int c0 = (threadIdx.x & (cvec - 1)) * 8;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    sc[j] = scale[c0 + j];
+    sh[j] = shift[c0 + j];
+  }
+  for (int r = blockIdx.x; r < NROWS; r += gridDim.x) {
+    const int n = r / Hp;                 // one scalar divide per row
+    const int yy = r - n * Hp;
+    const bool rrow = yy < ring || yy >= Hp - ring;
+    const long base = ((long)r * Wp) * cvec;   // row base in vec units
+    for (int j = threadIdx.x; j < rowvecs; j += BLOCK) {
+      const long i = base + j;
+      const int xpix = j >> cvec_sh;
+      if (rrow || xpix < ring || xpix >= Wp - ring) {
+        *reinterpret_cast<bf16x8*>(y + i * 8) = bf16x8{};
+        continue;
+      }
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(x + i * 8);
+      bf16x8 o;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float vv = (float)v[k] * sc[k] + sh[k];
+        o[k] = (__bf16)act_fwd(vv, act);
+      }
+      *reinterpret_cast<bf16x8*>(y + i * 8) = o;
+    }
   }
 }
 
@@ -165,7 +200,7 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     float* __restrict__ rows,  // (gridDim.x, 2, C) per-block partials (stores)
-    long nvec, int C, int act, int Hp, int Wp, int ring) {
+    long nvec, int C, int act) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float p1[8], p2[8];
@@ -182,14 +217,6 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
 
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    if (ring > 0) {
-      const long pix = i / cvec;
-      const long rem = pix % ((long)Hp * Wp);
-      const int yy = (int)(rem / Wp);
-      const int xx = (int)(rem - (long)yy * Wp);
-      if (yy < ring || yy >= Hp - ring || xx < ring || xx >= Wp - ring)
-        continue;   // x ring is garbage, dy ring grads are for constants
-    }
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
 #pragma unroll
@@ -233,6 +260,128 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_kernel(
 #pragma unroll
     for (int w = 0; w < BLOCK / 64; ++w) t += sred[w * 2 * C + i];
     row[i] = t;
+  }
+}
+
+// ring row-walk reduce: block grid-strides over padded rows, skipping ring
+// positions with shift/mask tests only; per-block partial row stores (same
+// deterministic fold/combine as the dense path).
+__global__ __launch_bounds__(BLOCK) void bn_act_bwd_reduce_ring_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ rows, int NROWS, int Hp, int Wp, int C, int act,
+    int ring, int cvec_sh) {
+  const int cvec = 1 << cvec_sh;
+  const int rowvecs = Wp * C / 8;
+  const int c0 = (threadIdx.x & (cvec - 1)) * 8;
+  float p1[8], p2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) p1[j] = p2[j] = 0.f;
+  float mn[8], is[8], ga[8], be[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mn[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+    ga[j] = gamma[c0 + j];
+    be[j] = beta[c0 + j];
+  }
+  for (int r = blockIdx.x; r < NROWS; r += gridDim.x) {
+    const int n = r / Hp;
+    const int yy = r - n * Hp;
+    if (yy < ring || yy >= Hp - ring) continue;
+    const long base = ((long)r * Wp) * cvec;
+    for (int j = threadIdx.x; j < rowvecs; j += BLOCK) {
+      const int xpix = j >> cvec_sh;
+      if (xpix < ring || xpix >= Wp - ring) continue;
+      const long i = base + j;
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
+      bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float xhat = ((float)xv[k] - mn[k]) * is[k];
+        const float dyp =
+            (float)gv[k] * act_bwd_from_u(ga[k] * xhat + be[k], act);
+        p1[k] += dyp;
+        p2[k] += dyp * xhat;
+      }
+    }
+  }
+  const int lane = threadIdx.x & 63;
+  extern __shared__ float sred[];
+  const int wave = threadIdx.x >> 6;
+  if (cvec < 64) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      for (int off = 32; off >= cvec; off >>= 1) {
+        p1[j] += __shfl_down(p1[j], off, 64);
+        p2[j] += __shfl_down(p2[j], off, 64);
+      }
+    }
+  }
+  if (lane < min(cvec, 64)) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      sred[wave * 2 * C + c0 + j] = p1[j];
+      sred[wave * 2 * C + C + c0 + j] = p2[j];
+    }
+  }
+  __syncthreads();
+  float* row = rows + (long)blockIdx.x * 2 * C;
+  for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
+    float t = 0.f;
+#pragma unroll
+    for (int w = 0; w < BLOCK / 64; ++w) t += sred[w * 2 * C + i];
+    row[i] = t;
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_ring_kernel(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    const float* __restrict__ scale, const float* __restrict__ red,
+    __bf16* __restrict__ dx, int NROWS, int Hp, int Wp, int C, int act,
+    float inv_count, int ring, int cvec_sh) {
+  const int cvec = 1 << cvec_sh;
+  const int rowvecs = Wp * C / 8;
+  const int c0 = (threadIdx.x & (cvec - 1)) * 8;
+  float mn[8], is[8], ga[8], be[8], sc[8], r1[8], r2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mn[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+    ga[j] = gamma[c0 + j];
+    be[j] = beta[c0 + j];
+    sc[j] = scale[c0 + j];
+    r1[j] = red[c0 + j] * inv_count;
+    r2[j] = red[C + c0 + j] * inv_count;
+  }
+  for (int r = blockIdx.x; r < NROWS; r += gridDim.x) {
+    const int n = r / Hp;
+    const int yy = r - n * Hp;
+    const bool rrow = yy < ring || yy >= Hp - ring;
+    const long base = ((long)r * Wp) * cvec;
+    for (int j = threadIdx.x; j < rowvecs; j += BLOCK) {
+      const long i = base + j;
+      const int xpix = j >> cvec_sh;
+      if (rrow || xpix < ring || xpix >= Wp - ring) {
+        // the previous conv's dgrad GATHERS dx's ring: must be zero
+        *reinterpret_cast<bf16x8*>(dx + i * 8) = bf16x8{};
+        continue;
+      }
+      bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
+      bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
+      bf16x8 o;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        const float xhat = ((float)xv[k] - mn[k]) * is[k];
+        const float dyp =
+            (float)gv[k] * act_bwd_from_u(ga[k] * xhat + be[k], act);
+        o[k] = (__bf16)(sc[k] * (dyp - r1[k] - xhat * r2[k]));
+      }
+      *reinterpret_cast<bf16x8*>(dx + i * 8) = o;
+    }
   }
 }
 
@@ -355,8 +504,7 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     const float* __restrict__ scale, const float* __restrict__ red,
-    __bf16* __restrict__ dx, long nvec, int C, int act, float inv_count,
-    int Hp, int Wp, int ring) {
+    __bf16* __restrict__ dx, long nvec, int C, int act, float inv_count) {
   const int cvec = C / 8;
   const int c0 = (int)(((long)blockIdx.x * BLOCK + threadIdx.x) % cvec) * 8;
   float mn[8], is[8], ga[8], be[8], sc[8], r1[8], r2[8];
@@ -372,17 +520,6 @@ __global__ __launch_bounds__(BLOCK) void bn_act_bwd_apply_kernel(
   }
   for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nvec;
        i += (long)gridDim.x * BLOCK) {
-    if (ring > 0) {
-      const long pix = i / cvec;
-      const long rem = pix % ((long)Hp * Wp);
-      const int yy = (int)(rem / Wp);
-      const int xx = (int)(rem - (long)yy * Wp);
-      if (yy < ring || yy >= Hp - ring || xx < ring || xx >= Wp - ring) {
-        // the previous conv's dgrad GATHERS dx's ring: must be zero
-        *reinterpret_cast<bf16x8*>(dx + i * 8) = bf16x8{};
-        continue;
-      }
-    }
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i * 8);
     bf16x8 gv = *reinterpret_cast<const bf16x8*>(dy + i * 8);
     bf16x8 o;
@@ -475,11 +612,22 @@ std::vector<torch::Tensor> bn_act_fwd_train(
                      (float)momentum, (float)eps, (float)count, C, nbuckets);
 
   const long nvec = x.numel() / 8;
-  hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK), 0,
-                     stream, reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     scale.data_ptr<float>(), shift.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C, (int)act,
-                     Hp, Wp, (int)ring);
+  if (ring > 0) {
+    const int NROWS = (int)x.size(0) * Hp;
+    hipLaunchKernelGGL(bn_act_fwd_ring_kernel,
+                       dim3(std::min(2048, NROWS)), dim3(BLOCK), 0, stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(y.data_ptr()), NROWS, Hp, Wp,
+                       C, (int)act, (int)ring, __builtin_ctz(C / 8));
+  } else {
+    hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK),
+                       0, stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C,
+                       (int)act, Hp, Wp, 0);
+  }
   return {y, mean, invstd, scale};
 }
 
@@ -506,11 +654,22 @@ torch::Tensor bn_act_fwd_eval(torch::Tensor x, torch::Tensor gamma,
                      invstd.data_ptr<float>(), scale.data_ptr<float>(),
                      shift.data_ptr<float>(), (float)eps, C);
   const long nvec = x.numel() / 8;
-  hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK), 0,
-                     stream, reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     scale.data_ptr<float>(), shift.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C, (int)act,
-                     Hp, Wp, (int)ring);
+  if (ring > 0) {
+    const int NROWS = (int)x.size(0) * Hp;
+    hipLaunchKernelGGL(bn_act_fwd_ring_kernel,
+                       dim3(std::min(2048, NROWS)), dim3(BLOCK), 0, stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(y.data_ptr()), NROWS, Hp, Wp,
+                       C, (int)act, (int)ring, __builtin_ctz(C / 8));
+  } else {
+    hipLaunchKernelGGL(bn_act_fwd_kernel, dim3(pick_grid(nvec)), dim3(BLOCK),
+                       0, stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(y.data_ptr()), nvec, C,
+                       (int)act, Hp, Wp, 0);
+  }
   return y;
 }
 
@@ -534,17 +693,29 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
   auto dx = torch::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   const long nvec = x.numel() / 8;
-  const int rgrid = pick_grid(nvec);
+  const int NROWS = (int)x.size(0) * Hp;
+  const int rgrid =
+      ring > 0 ? std::min(2048, NROWS) : pick_grid(nvec);
   auto rows = torch::empty({rgrid, 2, C}, f32);     // stores — never zeroed
   auto red = torch::empty({2, C}, f32);
-  hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(rgrid), dim3(BLOCK),
-                     (BLOCK / 64) * 2 * C * sizeof(float), stream,
-                     reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     reinterpret_cast<const __bf16*>(dy.data_ptr()),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                     rows.data_ptr<float>(), nvec, C, (int)act, Hp, Wp,
-                     (int)ring);
+  if (ring > 0) {
+    hipLaunchKernelGGL(bn_act_bwd_reduce_ring_kernel, dim3(rgrid), dim3(BLOCK),
+                       (BLOCK / 64) * 2 * C * sizeof(float), stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       reinterpret_cast<const __bf16*>(dy.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       rows.data_ptr<float>(), NROWS, Hp, Wp, C, (int)act,
+                       (int)ring, __builtin_ctz(C / 8));
+  } else {
+    hipLaunchKernelGGL(bn_act_bwd_reduce_kernel, dim3(rgrid), dim3(BLOCK),
+                       (BLOCK / 64) * 2 * C * sizeof(float), stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       reinterpret_cast<const __bf16*>(dy.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       rows.data_ptr<float>(), nvec, C, (int)act);
+  }
   const int accumulate = dgamma_acc.has_value() ? 1 : 0;
   torch::Tensor dgamma =
       accumulate ? dgamma_acc.value() : torch::empty({C}, f32);
@@ -559,15 +730,28 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                      dim3(256), 0, stream, folded.data_ptr<float>(), nrows,
                      red.data_ptr<float>(), dgamma.data_ptr<float>(),
                      dbeta.data_ptr<float>(), C, accumulate);
-  hipLaunchKernelGGL(bn_act_bwd_apply_kernel, dim3(pick_grid(nvec)),
-                     dim3(BLOCK), 0, stream,
-                     reinterpret_cast<const __bf16*>(x.data_ptr()),
-                     reinterpret_cast<const __bf16*>(dy.data_ptr()),
-                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                     scale.data_ptr<float>(), red.data_ptr<float>(),
-                     reinterpret_cast<__bf16*>(dx.data_ptr()), nvec, C,
-                     (int)act, (float)(1.0 / count), Hp, Wp, (int)ring);
+  if (ring > 0) {
+    hipLaunchKernelGGL(bn_act_bwd_apply_ring_kernel,
+                       dim3(std::min(2048, NROWS)), dim3(BLOCK), 0, stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       reinterpret_cast<const __bf16*>(dy.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       scale.data_ptr<float>(), red.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), NROWS, Hp, Wp,
+                       C, (int)act, (float)(1.0 / count), (int)ring,
+                       __builtin_ctz(C / 8));
+  } else {
+    hipLaunchKernelGGL(bn_act_bwd_apply_kernel, dim3(pick_grid(nvec)),
+                       dim3(BLOCK), 0, stream,
+                       reinterpret_cast<const __bf16*>(x.data_ptr()),
+                       reinterpret_cast<const __bf16*>(dy.data_ptr()),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       scale.data_ptr<float>(), red.data_ptr<float>(),
+                       reinterpret_cast<__bf16*>(dx.data_ptr()), nvec, C,
+                       (int)act, (float)(1.0 / count));
+  }
   return {dx, dgamma, dbeta};
 }
 

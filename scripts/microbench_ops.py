@@ -69,6 +69,27 @@ for tag, C, H, K in SHAPES:
     report(f"{tag} bn_act_bwd (acc)",
            timeit(lambda: ext.bn_act_bwd(out, g, mean, invstd, gamma, beta, scale, 1,
                                          gamma.clone(), beta.clone())))
+    # padded (ring=1) variants: conv writes a padded map, bn walks rows
+    outp, statsp = ext.conv2d_nhwc_fwd(x, w, None, 1, 1, 0, True,
+                                       H + 2, H + 2, 1, 1)
+    gp = mk(B, K, H + 2, H + 2)
+    report(f"{tag} conv fwd padded-out",
+           timeit(lambda: ext.conv2d_nhwc_fwd(x, w, None, 1, 1, 0, True,
+                                              H + 2, H + 2, 1, 1)))
+    report(f"{tag} bn_act_fwd ring=1",
+           timeit(lambda: ext.bn_act_fwd_train(outp, statsp, gamma, beta, rm, rv,
+                                               0.1, 1e-5, 1, 1)))
+    yp, meanp, invstdp, scalep = ext.bn_act_fwd_train(outp, statsp, gamma, beta,
+                                                      rm, rv, 0.1, 1e-5, 1, 1)
+    report(f"{tag} bn_act_bwd ring=1",
+           timeit(lambda: ext.bn_act_bwd(outp, gp, meanp, invstdp, gamma, beta,
+                                         scalep, 1, None, None, 1)))
+    xp = mk(B, C, H + 2, H + 2)
+    report(f"{tag} glds fwd (padded in+out)",
+           timeit(lambda: ext.conv2d_glds_fwd(xp, w, None, 1, 0, True,
+                                              H + 2, H + 2, 1, 1)))
+    report(f"{tag} wgrad yring=1",
+           timeit(lambda: ext.conv2d_nhwc_wgrad(gp, xp, 3, 3, 1, 0, 0, None, 1)))
     report(f"{tag} wgrad",
            timeit(lambda: ext.conv2d_nhwc_wgrad(g, x, 3, 3, 1, 1, 0)))
     report(f"{tag} wgrad (acc)",
