@@ -149,3 +149,36 @@ def test_managed_grads_match_autograd_accumulation():
         s = b.abs().max().item() + 1e-6
         e = (a - b).abs().max().item()
         assert e <= 1e-4 * s + 1e-5, f"{n}: managed vs autograd err {e} (scale {s})"
+
+
+@pytest.mark.parametrize("n,c,h,k,ks,st,yr", [
+    (4, 64, 32, 64, 3, 1, 1),      # glds path, ringed Y
+    (4, 128, 16, 256, 3, 1, 0),    # glds path, dense Y
+    (3, 64, 32, 128, 4, 2, 1),     # k4s2 ringed
+    (2, 64, 20, 128, 3, 1, 1),     # non-pow2 spatial -> register path
+])
+def test_wgrad_padded_matches_reference(n, c, h, k, ks, st, yr):
+    """wgrad with padded operands (PAD=0, X in-bounds, Y interior-mapped)
+    equals torch's conv2d_weight on the dense tensors — covers both the
+    glds-staged and the register-staged fallback."""
+    import torch.nn.functional as F
+    from p2pvg_amd.ops import _hip_ext_loader
+
+    ext = _hip_ext_loader.load()
+    torch.manual_seed(0)
+    pad = 1
+    x = torch.randn(n, c, h, h, device="cuda")
+    ho = (h + 2 * pad - ks) // st + 1
+    g = torch.randn(n, k, ho, ho, device="cuda")
+    xb = x.bfloat16().float()
+    gb = g.bfloat16().float()
+    ref = torch.nn.grad.conv2d_weight(
+        xb, (k, c, ks, ks), gb, stride=st, padding=pad)
+
+    xp = F.pad(x, (pad,) * 4).bfloat16().contiguous(memory_format=CL)
+    gp = (F.pad(g, (yr,) * 4) if yr else g).bfloat16().contiguous(memory_format=CL)
+    ws = ext.conv2d_nhwc_wgrad(gp, xp, ks, ks, st, 0, 0, None, yr)
+    got = ws.permute(0, 3, 1, 2)  # (K, ks, ks, C) -> (K, C, ks, ks)
+    scale = ref.abs().max().item() + 1e-6
+    err = (got.float() - ref).abs().max().item()
+    assert err < 2e-2 * scale + 2e-2, f"wgrad err {err} scale {scale}"
