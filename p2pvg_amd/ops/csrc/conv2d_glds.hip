@@ -186,8 +186,13 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
     const int buf = t % 3;
     const char* abuf = lds + buf * SLAB;
     const char* bbuf = abuf + ABYTES;
-    // chunk t landed once <= 2 chunks' glds remain outstanding
-    waitcnt_vm<2 * G>();
+    // chunk t landed once its own glds completed; with fewer than 2 chunks
+    // staged ahead (short K-loops / loop tail), a flat vmcnt(2G) would pass
+    // while chunk t is still in flight
+    const int ahead = min(nchunks, t + 3) - 1 - t;
+    if (ahead >= 2) waitcnt_vm<2 * G>();
+    else if (ahead == 1) waitcnt_vm<G>();
+    else waitcnt_vm<0>();
     barrier_mem();
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {

@@ -397,7 +397,12 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_glds_kernel(
 
   for (int t = 0; t < nfull; ++t) {
     const int buf = t % 3;
-    wwaitcnt_vm<2 * G>();
+    // wait until chunk t's own glds completed: with fewer than 3 chunks
+    // staged ahead, vmcnt(2G) would pass while chunk t is still in flight
+    const int ahead = min(nfull, t + 3) - 1 - t;
+    if (ahead >= 2) wwaitcnt_vm<2 * G>();
+    else if (ahead == 1) wwaitcnt_vm<G>();
+    else wwaitcnt_vm<0>();
     wbarrier_mem();
     mfma_image(wlds + buf * SLAB, wlds + buf * SLAB + YBYTES);
     wbarrier_mem();
