@@ -29,6 +29,21 @@ def _cat_skip(a, b):
     return r
 
 
+def _skip_join(a, b):
+    """Hand (up_out, skip) to the consuming conv as a PAIR when the
+    dual-pointer glds path can gather both directly (SURVEY K8 concat
+    elimination — no cat kernel, no cat backward splits); falls back to the
+    marked concat otherwise (CPU, torch backend, odd channel splits). The
+    first fused conv of the receiving FusedSequential consumes the pair."""
+    from ...ops.conv import _use_hip_path
+
+    if (getattr(a, "_pvg_pad", 0) and _use_hip_path(a)
+            and a.shape[1] % 64 == 0 and b.shape[1] % 64 == 0
+            and getattr(a, "_pvg_pad", 0) == getattr(b, "_pvg_pad", 0)):
+        return (a, b)
+    return _cat_skip(a, b)
+
+
 class Encoder64(nn.Module):
     def __init__(self, dim: int, nc: int = 1):
         super().__init__()
@@ -83,10 +98,10 @@ class Decoder64(nn.Module):
     def forward(self, inp):
         vec, skip = inp
         d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
-        d2 = self.upc2(_cat_skip(self.up(d1, pad_out=True), skip[3]))
-        d3 = self.upc3(_cat_skip(self.up(d2, pad_out=True), skip[2]))
-        d4 = self.upc4(_cat_skip(self.up(d3, pad_out=True), skip[1]))
-        return self.upc5(_cat_skip(self.up(d4, pad_out=True), skip[0]))
+        d2 = self.upc2(_skip_join(self.up(d1, pad_out=True), skip[3]))
+        d3 = self.upc3(_skip_join(self.up(d2, pad_out=True), skip[2]))
+        d4 = self.upc4(_skip_join(self.up(d3, pad_out=True), skip[1]))
+        return self.upc5(_skip_join(self.up(d4, pad_out=True), skip[0]))
 
 
 class Encoder128(nn.Module):
@@ -150,8 +165,8 @@ class Decoder128(nn.Module):
     def forward(self, inp):
         vec, skip = inp
         d1 = self.upc1(vec.view(-1, self.dim, 1, 1))
-        d2 = self.upc2(_cat_skip(self.up(d1, pad_out=True), skip[4]))
-        d3 = self.upc3(_cat_skip(self.up(d2, pad_out=True), skip[3]))
-        d4 = self.upc4(_cat_skip(self.up(d3, pad_out=True), skip[2]))
-        d5 = self.upc5(_cat_skip(self.up(d4, pad_out=True), skip[1]))
-        return self.upc6(_cat_skip(self.up(d5, pad_out=True), skip[0]))
+        d2 = self.upc2(_skip_join(self.up(d1, pad_out=True), skip[4]))
+        d3 = self.upc3(_skip_join(self.up(d2, pad_out=True), skip[3]))
+        d4 = self.upc4(_skip_join(self.up(d3, pad_out=True), skip[2]))
+        d5 = self.upc5(_skip_join(self.up(d4, pad_out=True), skip[1]))
+        return self.upc6(_skip_join(self.up(d5, pad_out=True), skip[0]))

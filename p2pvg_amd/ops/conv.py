@@ -327,6 +327,86 @@ class Conv2dNHWCFn(torch.autograd.Function):
         return (dx, dw, db) + (None,) * 8
 
 
+class CatConv2dFn(torch.autograd.Function):
+    """y = conv2d(cat([x1, x2], 1), w) WITHOUT materializing the concat
+    (SURVEY §2.6 K8): the glds conv gathers from two source tensors (the
+    chunk's channel picks the pointer — every skip split is 64-aligned),
+    its dgrad scatters into two grad tensors (dual-destination epilogue),
+    and the wgrad gathers dual-X. Requires the padded-ring path (both
+    inputs padded by the conv's padding) and the managed-grad flow for w."""
+
+    @staticmethod
+    def forward(ctx, x1, x2, w, stride: int, pad: int, act: int,
+                want_stats: bool, w_fwd, w_bwd, ring: int, out_ring: int):
+        ext = _ext()
+        wf = w_fwd if w_fwd is not None else w
+        k = wf.shape[2]
+        assert ring == pad and ring > 0, "CatConv2dFn needs the padded path"
+        oh = ow = oy = 0
+        if out_ring:
+            H = x1.shape[2] - 2 * ring
+            ho = (H + 2 * pad - k) // stride + 1
+            oh = ow = ho + 2 * out_ring
+            oy = out_ring
+        out, stats = ext.conv2d_glds_fwd(x1, wf, None, stride, act,
+                                         want_stats, oh, ow, oy, oy, x2)
+        wb = w_bwd if w_bwd is not None else wf
+        ctx.save_for_backward(x1, x2, wb)
+        ctx.shadow_bwd = w_bwd is not None
+        ctx.meta = (stride, pad, act, ring, out_ring, k)
+        ctx.wref = w
+        ctx.wdtype = w.dtype
+        if stats is None or not want_stats:
+            stats = torch.empty(0, device=x1.device)
+        ctx.mark_non_differentiable(stats)
+        return out, stats
+
+    @staticmethod
+    def backward(ctx, gout, _gstats):
+        x1, x2, w = ctx.saved_tensors
+        stride, pad, act, ring, out_ring, k = ctx.meta
+        ext = _ext()
+        gout = gout.contiguous(memory_format=CL)
+        if gout.dtype != torch.bfloat16:
+            gout = gout.to(torch.bfloat16)
+        wge = weight_grads_enabled()
+        dx1 = dx2 = dw = None
+        if ctx.needs_input_grad[0] or ctx.needs_input_grad[1]:
+            wt = w if ctx.shadow_bwd else \
+                w.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)
+            dx2 = torch.empty_like(x2)
+            dx1, _ = ext.conv2d_glds_fwd(gout, wt, None, 1,
+                                         0, False, x1.shape[2], x1.shape[3],
+                                         ring, ring, None, dx2)
+        if ctx.needs_input_grad[2] and wge:
+            wg = _acc_target(ctx.wref)
+            if wg is not None and ctx.wdtype == torch.float32 \
+                    and wg.is_contiguous(memory_format=CL):
+                ext.conv2d_nhwc_wgrad(gout, x1, k, k, stride, 0, 0, wg,
+                                      out_ring, x2)
+            else:
+                ws = ext.conv2d_nhwc_wgrad(gout, x1, k, k, stride, 0, 0,
+                                           None, out_ring, x2)
+                dw = ws.permute(0, 3, 1, 2)
+                if ctx.wdtype != torch.float32:
+                    dw = dw.to(torch.bfloat16).contiguous(memory_format=CL)
+        return (dx1, dx2, dw) + (None,) * 8
+
+
+def cat_conv_eligible(x1, x2, conv) -> bool:
+    """Can conv consume (x1, x2) via the dual-pointer glds path?"""
+    ring = getattr(x1, "_pvg_pad", 0)
+    if ring == 0 or ring != getattr(x2, "_pvg_pad", 0):
+        return False
+    if not isinstance(conv, nn.Conv2d) or isinstance(conv, nn.ConvTranspose2d):
+        return False
+    C = x1.shape[1] + x2.shape[1]
+    return (conv.kernel_size[0] == 3 and conv.stride[0] == 1
+            and conv.padding[0] == ring and x1.shape[1] % 64 == 0
+            and x2.shape[1] % 64 == 0 and conv.out_channels % 64 == 0
+            and conv.in_channels == C and x1.shape[2:] == x2.shape[2:])
+
+
 class ConvT2dNHWCFn(torch.autograd.Function):
     """y = conv_transpose2d(x, w, b, stride, pad); w logical (Ci, Co, k, k)."""
 

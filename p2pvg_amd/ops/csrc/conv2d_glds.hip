@@ -70,6 +70,15 @@ struct GArgs {
   int act;
   int mblocks;
   float* stats;           // (mblocks*2, 2, K) per-(block, wave-row) stores
+  // dual-source input (skip-concat elimination): channels
+  // [0, C1) read from `in`, [C1, C) from `in2` (same N/H/W geometry).
+  // C1 % 64 == 0 so a BK=64 chunk never straddles the seam.
+  const __bf16* in2;
+  int C1;
+  // dual-destination output (the dgrad of a cat-consuming conv): columns
+  // [0, K1) write to `out`, [K1, K) to `out2` (same N/OH/OW geometry).
+  __bf16* out2;
+  int K1;
 };
 
 // KSIZE/STRIDE compile-time; BN: 64 or 128.
@@ -104,11 +113,11 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
       const int rem = pix - n * (a.HO * a.WO);
       const int ho = rem / a.WO;
       const int wo = rem - ho * a.WO;
-      // tap-(0,0) gather origin; ALL taps in-bounds by contract
-      pix_base[i] = ((long)n * a.H + ho * STRIDE) * a.W * a.C +
-                    (long)(wo * STRIDE) * a.C;
-      pix_out[i] = (((long)n * a.OH + (ho + a.oy0)) * a.OW + (wo + a.ox0)) *
-                   a.K;
+      // tap-(0,0) gather origin in PIXELS (channel stride applied at stage
+      // time: the two dual-source tensors have different C strides)
+      pix_base[i] = ((long)n * a.H + ho * STRIDE) * a.W +
+                    (long)(wo * STRIDE);
+      pix_out[i] = ((long)n * a.OH + (ho + a.oy0)) * a.OW + (wo + a.ox0);
     } else {
       pix_base[i] = 0;      // safe in-bounds dummy reads
       pix_out[i] = -1;
@@ -126,7 +135,7 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
   constexpr int ACALLS = GBM / 8 / 8;          // 4: rows per wave / 8
   constexpr int BCALLS = BN / 8 / 8;           // 1 or 2
   constexpr int BROWS = BN / 8;                // B rows per wave
-  long asrc[ACALLS];
+  int asrc[ACALLS];
   int arowl[ACALLS];
 #pragma unroll
   for (int i = 0; i < ACALLS; ++i) {
@@ -140,9 +149,9 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
     const int brow = wv * BROWS + i * 8 + rsub;
     bsrc[i] = (long)(k0 + brow) * RSC + ((piece ^ (brow & 7)) << 3);
   }
-  long abase[ACALLS];
+  int abase[ACALLS];
 #pragma unroll
-  for (int i = 0; i < ACALLS; ++i) abase[i] = pix_base[arowl[i]] + asrc[i];
+  for (int i = 0; i < ACALLS; ++i) abase[i] = (int)pix_base[arowl[i]];
 
   const int n_inner = a.C >> 6;                 // C / 64 chunks per tap
   const int nchunks = KSIZE * KSIZE * n_inner;
@@ -151,13 +160,19 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
     const int tap = t / n_inner;
     const int c0 = (t - tap * n_inner) << 6;
     const int ro = tap / KSIZE, so = tap % KSIZE;
-    const long atap = ((long)ro * a.W + so) * a.C + c0;
+    // dual-source: the whole 64-wide chunk comes from one tensor
+    const bool second = a.in2 != nullptr && c0 >= a.C1;
+    const __bf16* src = second ? a.in2 : in;
+    const int cs = a.in2 == nullptr ? a.C
+                                    : (second ? a.C - a.C1 : a.C1);
+    const int cl = second ? c0 - a.C1 : c0;
+    const int tappix = ro * a.W + so;
     const long btap = (long)tap * a.C + c0;
     char* abuf = lds + buf * SLAB;
     char* bbuf = abuf + ABYTES;
 #pragma unroll
     for (int i = 0; i < ACALLS; ++i) {
-      glds16(in + abase[i] + atap,
+      glds16(src + (long)(abase[i] + tappix) * cs + cl + asrc[i],
              abuf + (wv * (GBM / 8) + i * 8) * 128);
     }
 #pragma unroll
@@ -224,16 +239,22 @@ __global__ __launch_bounds__(GTHREADS) void conv2d_glds_kernel(
   for (int j = 0; j < FN; ++j) {
     const int col = k0 + wn + j * 16 + (lane & 15);
     const float bv = bias != nullptr ? bias[col] : 0.f;
+    // dual-destination: column decides the target tensor (dgrad of a
+    // cat-consuming conv splits back into the two operand grads)
+    const bool osecond = a.out2 != nullptr && col >= a.K1;
+    __bf16* dst = osecond ? a.out2 : out;
+    const int ks = a.out2 == nullptr ? a.K : (osecond ? a.K - a.K1 : a.K1);
+    const int ocol = osecond ? col - a.K1 : col;
     float csum = 0.f, csq = 0.f;
 #pragma unroll
     for (int i = 0; i < FM; ++i) {
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
         const int lrow = wm + i * 16 + (lane >> 4) * 4 + v;
-        const long ooff = pix_out[lrow];
-        if (ooff >= 0) {
+        const long opix = pix_out[lrow];
+        if (opix >= 0) {
           const float val = gactivate(acc[i][j][v] + bv, a.act);
-          out[ooff + col] = (__bf16)val;
+          dst[opix * ks + ocol] = (__bf16)val;
           csum += val;
           csq += val * val;
         }
@@ -304,15 +325,27 @@ std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
                                            c10::optional<torch::Tensor> bias,
                                            long stride, long act,
                                            bool want_stats, long oh, long ow,
-                                           long oy0, long ox0) {
+                                           long oy0, long ox0,
+                                           c10::optional<torch::Tensor> in2,
+                                           c10::optional<torch::Tensor> out2_k1) {
   TORCH_CHECK(in.is_cuda() && in.scalar_type() == torch::kBFloat16 &&
                   in.is_contiguous(at::MemoryFormat::ChannelsLast),
               "glds conv: in must be bf16 channels_last GPU");
   TORCH_CHECK(w.is_cuda() && w.scalar_type() == torch::kBFloat16 &&
                   w.is_contiguous(at::MemoryFormat::ChannelsLast),
               "glds conv: w must be bf16 channels_last GPU");
-  const int Nb = in.size(0), C = in.size(1), H = in.size(2), W = in.size(3);
+  const int Nb = in.size(0);
+  const int C1 = in.size(1);
+  const int C = in2.has_value() ? C1 + (int)in2->size(1) : C1;
+  const int H = in.size(2), W = in.size(3);
   const int K = w.size(0), R = w.size(2);
+  if (in2.has_value()) {
+    TORCH_CHECK(in2->is_cuda() && in2->scalar_type() == torch::kBFloat16 &&
+                    in2->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    in2->size(0) == Nb && in2->size(2) == H &&
+                    in2->size(3) == W && C1 % 64 == 0,
+                "glds conv: in2 geometry mismatch (C1 must be 64-aligned)");
+  }
   TORCH_CHECK(w.size(1) == C && w.size(3) == R, "glds conv: weight mismatch");
   TORCH_CHECK(conv2d_glds_eligible(C, K, R, stride, 0),
               "glds conv: unsupported geometry");
@@ -321,8 +354,10 @@ std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
   const int OH = oh > 0 ? (int)oh : HO;
   const int OW = ow > 0 ? (int)ow : WO;
 
+  const int Kout = out2_k1.has_value() ? K - (int)out2_k1->size(1) : K;
   auto out = torch::empty(
-      {Nb, K, OH, OW}, in.options().memory_format(at::MemoryFormat::ChannelsLast));
+      {Nb, Kout, OH, OW},
+      in.options().memory_format(at::MemoryFormat::ChannelsLast));
   if (bias.has_value()) {
     CHECK_INPUT(bias.value());
     TORCH_CHECK(bias->scalar_type() == torch::kFloat32, "bias must be fp32");
@@ -332,6 +367,22 @@ std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
   a.HO = HO; a.WO = WO; a.OH = OH; a.OW = OW;
   a.oy0 = (int)oy0; a.ox0 = (int)ox0;
   a.act = (int)act;
+  a.in2 = in2.has_value()
+              ? reinterpret_cast<const __bf16*>(in2->data_ptr()) : nullptr;
+  a.C1 = C1;
+  a.out2 = nullptr;
+  a.K1 = K;
+  if (out2_k1.has_value()) {
+    auto& o2 = out2_k1.value();
+    TORCH_CHECK(o2.is_cuda() && o2.scalar_type() == torch::kBFloat16 &&
+                    o2.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    o2.size(0) == Nb && (int)o2.size(2) == OH &&
+                    (int)o2.size(3) == OW,
+                "glds conv: out2 geometry mismatch");
+    a.out2 = reinterpret_cast<__bf16*>(o2.data_ptr());
+    a.K1 = K - (int)o2.size(1);
+    TORCH_CHECK(a.K1 > 0 && a.K1 < K, "glds conv: bad out2 split");
+  }
   torch::Tensor stats_out;
 
   const bool bn64 = (K % 128 != 0);

@@ -30,11 +30,11 @@
 #include <cstdlib>
 
 // glds 3-buffer pipeline for the in-bounds (pad=0 / padded-input) case
-std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
-                                           c10::optional<torch::Tensor> bias,
-                                           long stride, long act,
-                                           bool want_stats, long oh, long ow,
-                                           long oy0, long ox0);
+std::vector<torch::Tensor> conv2d_glds_fwd(
+    torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    long stride, long act, bool want_stats, long oh, long ow, long oy0,
+    long ox0, c10::optional<torch::Tensor> in2,
+    c10::optional<torch::Tensor> out2_k1);
 bool conv2d_glds_eligible(long C, long K, long R, long stride, long pad);
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
@@ -465,7 +465,7 @@ std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
       ((long)Nb * HO * WO / 256) * (K / (K % 128 ? 64 : 128)) >= 200 &&
       glds_enabled()) {
     return conv2d_glds_fwd(in, w, bias, stride, act, want_stats, oh, ow, oy0,
-                           ox0);
+                           ox0, c10::nullopt, c10::nullopt);
   }
 
   const int OH = oh > 0 ? (int)oh : HO;

@@ -278,7 +278,7 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_glds_kernel(
     const __bf16* __restrict__ Y, const __bf16* __restrict__ X,
     float* __restrict__ ws, int Nb, int HO, int WO, int B, int H, int W,
     int A, int R, int S, int STRIDE, int p_per_slab, int yring,
-    int howo_sh, int wo_sh) {
+    int howo_sh, int wo_sh, const __bf16* __restrict__ X2, int A1) {
   extern __shared__ __align__(16) char wlds[];
   constexpr int YBYTES = PCH * BTB * 2;
   constexpr int XBYTES = PCH * BTA * 2;
@@ -351,9 +351,14 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_glds_kernel(
       const int wo = rem - (ho << wo_sh);
       const int hi = ho * STRIDE + r;     // PAD == 0, in-bounds by contract
       const int wi = wo * STRIDE + s;
-      const long xoff =
-          ((long)n * H * W + (long)hi * W + wi) * A + a0 + ch0;
-      wglds16(X + xoff, ximg + (wid * XC + i) * 1024);
+      // dual-X (skip-concat elimination): channel picks the source tensor
+      const int ach = a0 + ch0;
+      const bool second = X2 != nullptr && ach >= A1;
+      const __bf16* xsrc = second ? X2 : X;
+      const int as = X2 == nullptr ? A : (second ? A - A1 : A1);
+      const int al = second ? ach - A1 : ach;
+      const long xoff = ((long)n * H * W + (long)hi * W + wi) * as + al;
+      wglds16(xsrc + xoff, ximg + (wid * XC + i) * 1024);
     }
   };
 
@@ -450,8 +455,13 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_glds_kernel(
         const int wo = rem - (ho << wo_sh);
         const int hi = ho * STRIDE + r;
         const int wi = wo * STRIDE + s;
+        const int ach = a0 + ch0;
+        const bool second = X2 != nullptr && ach >= A1;
+        const __bf16* xsrc = second ? X2 : X;
+        const int as = X2 == nullptr ? A : (second ? A - A1 : A1);
+        const int al = second ? ach - A1 : ach;
         vx = *reinterpret_cast<const bf16x8*>(
-            X + ((long)n * H * W + (long)hi * W + wi) * A + a0 + ch0);
+            xsrc + ((long)n * H * W + (long)hi * W + wi) * as + al);
       }
       *reinterpret_cast<bf16x8*>(ximg + sub_off<BTA>(pixl, ch0)) = vx;
     }
@@ -499,7 +509,8 @@ __global__ __launch_bounds__(256) void wgrad_combine_kernel(
 // accumulation adds entirely (the ~29 uses per weight per step land here).
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp,
-                                c10::optional<torch::Tensor> acc, long yring) {
+                                c10::optional<torch::Tensor> acc, long yring,
+                                c10::optional<torch::Tensor> X2) {
   TORCH_CHECK(Y.is_cuda() && Y.scalar_type() == torch::kBFloat16 &&
                   Y.is_contiguous(at::MemoryFormat::ChannelsLast),
               "Y must be bf16 channels_last GPU");
@@ -508,7 +519,16 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
               "X must be bf16 channels_last GPU");
   const int Nb = Y.size(0), B = Y.size(1);
   const int HO = Y.size(2) - 2 * (int)yring, WO = Y.size(3) - 2 * (int)yring;
-  const int A = X.size(1), H = X.size(2), W = X.size(3);
+  const int A1 = X.size(1);
+  const int A = X2.has_value() ? A1 + (int)X2->size(1) : A1;
+  const int H = X.size(2), W = X.size(3);
+  if (X2.has_value()) {
+    TORCH_CHECK(X2->is_cuda() && X2->scalar_type() == torch::kBFloat16 &&
+                    X2->is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                    X2->size(0) == Nb && X2->size(2) == H &&
+                    X2->size(3) == W && A1 % 64 == 0,
+                "wgrad: X2 geometry mismatch");
+  }
   TORCH_CHECK(X.size(0) == Nb, "batch mismatch");
   const long E = (long)B * R * S * A;
 
@@ -535,6 +555,8 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
   const bool pow2 = ((HO * WO) & (HO * WO - 1)) == 0 && (WO & (WO - 1)) == 0;
   const bool use_glds = pad == 0 && pow2 && HO > 0 && WO > 0 &&
                         B % BTB == 0 && A % BTA == 0 && p_total >= 4 * PCH;
+  TORCH_CHECK(!X2.has_value() || use_glds,
+              "wgrad: dual-X requires the glds-eligible geometry");
   if (use_glds) {
     // PAD==0 means every X gather is in-bounds (padded operand or genuine
     // valid conv): glds 3-buffer pipeline, shift-only pixel decode
@@ -557,7 +579,11 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                        reinterpret_cast<const __bf16*>(X.data_ptr()),         \
                        ws.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,  \
                        (int)S, (int)stride, p_per_slab, (int)yring, howo_sh,  \
-                       wo_sh);                                                \
+                       wo_sh,                                                 \
+                       X2.has_value()                                         \
+                           ? reinterpret_cast<const __bf16*>(X2->data_ptr())  \
+                           : nullptr,                                         \
+                       A1);                                                   \
   } while (0)
     if (BTB == 128 && BTA == 128) WGRAD_GLAUNCH(128, 128);
     else if (BTB == 128) WGRAD_GLAUNCH(128, 64);

@@ -34,11 +34,11 @@ std::vector<torch::Tensor> conv2d_nhwc_fracstride(
     torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
     long up_stride, long up_pad, long OH, long OW, long act, bool want_stats,
     long in_ring, long out_ring);
-std::vector<torch::Tensor> conv2d_glds_fwd(torch::Tensor in, torch::Tensor w,
-                                           c10::optional<torch::Tensor> bias,
-                                           long stride, long act,
-                                           bool want_stats, long oh, long ow,
-                                           long oy0, long ox0);
+std::vector<torch::Tensor> conv2d_glds_fwd(
+    torch::Tensor in, torch::Tensor w, c10::optional<torch::Tensor> bias,
+    long stride, long act, bool want_stats, long oh, long ow, long oy0,
+    long ox0, c10::optional<torch::Tensor> in2,
+    c10::optional<torch::Tensor> out2_k1);
 std::vector<torch::Tensor> bn_act_fwd_train(
     torch::Tensor x, torch::Tensor stats, torch::Tensor gamma,
     torch::Tensor beta, c10::optional<torch::Tensor> running_mean,
@@ -57,7 +57,8 @@ std::vector<torch::Tensor> bn_act_bwd(torch::Tensor x, torch::Tensor dy,
                                       long ring);
 torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                                 long S, long stride, long pad, long splitp,
-                                c10::optional<torch::Tensor> acc, long yring);
+                                c10::optional<torch::Tensor> acc, long yring,
+                                c10::optional<torch::Tensor> X2);
 std::vector<torch::Tensor> maxpool2x2_fwd(torch::Tensor in, long ri, long ro);
 torch::Tensor maxpool2x2_bwd(torch::Tensor gout, torch::Tensor idx, long H,
                              long W, long ri, long ro);
@@ -90,7 +91,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("stride"), pybind11::arg("act"),
         pybind11::arg("want_stats") = false, pybind11::arg("oh") = 0,
         pybind11::arg("ow") = 0, pybind11::arg("oy0") = 0,
-        pybind11::arg("ox0") = 0);
+        pybind11::arg("ox0") = 0, pybind11::arg("in2") = c10::nullopt,
+        pybind11::arg("out2") = c10::nullopt);
   m.def("bn_act_fwd_train", &bn_act_fwd_train, "fused BN+act train fwd (gfx950)",
         pybind11::arg("x"), pybind11::arg("stats"), pybind11::arg("gamma"),
         pybind11::arg("beta"), pybind11::arg("running_mean"),
@@ -117,7 +119,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         pybind11::arg("Y"), pybind11::arg("X"), pybind11::arg("R"),
         pybind11::arg("S"), pybind11::arg("stride"), pybind11::arg("pad"),
         pybind11::arg("splitp") = 0, pybind11::arg("acc") = c10::nullopt,
-        pybind11::arg("yring") = 0);
+        pybind11::arg("yring") = 0, pybind11::arg("X2") = c10::nullopt);
   m.def("maxpool2x2_fwd", &maxpool2x2_fwd, "NHWC 2x2/s2 maxpool fwd (gfx950)",
         pybind11::arg("in"), pybind11::arg("ri") = 0, pybind11::arg("ro") = 0);
   m.def("maxpool2x2_bwd", &maxpool2x2_bwd, "NHWC 2x2/s2 maxpool bwd (gfx950)",

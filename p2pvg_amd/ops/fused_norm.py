@@ -96,26 +96,36 @@ def fused_conv_bn_act(x, conv, bn, act: int, pad_out: bool = False):
     padding (its gathers become in-bounds -> glds staging pipeline). An
     incoming padded x (attribute `_pvg_pad` set by the previous producer)
     routes the conv itself through the in-bounds path."""
-    from .conv import Conv2dNHWCFn, ConvT2dNHWCFn, _conv_shadows, _to_cl_bf16
+    from .conv import (CatConv2dFn, Conv2dNHWCFn, ConvT2dNHWCFn,
+                       _conv_shadows, _to_cl_bf16)
 
     ext = _ext()
     training = bn.training
-    in_ring = getattr(x, "_pvg_pad", 0)
+    dual = isinstance(x, tuple)
+    in_ring = getattr(x[0] if dual else x, "_pvg_pad", 0)
     out_ring = 1 if pad_out else 0
     with torch.autocast("cuda", enabled=False):
-        xl = _to_cl_bf16(x)
         sh = _conv_shadows(conv)
         want_stats = training
         # BatchNorm is shift-invariant, so the conv bias has EXACTLY zero
         # effect on the block output; skip it (and its gradient reduction).
         # The reference trains this bias on float-rounding noise only — it is
         # initialized to 0 (init_weights) and stays ~0.
-        if isinstance(conv, nn.ConvTranspose2d):
+        if dual:
+            # skip-concat elimination: the conv gathers from both sources
+            out, stats = CatConv2dFn.apply(
+                _to_cl_bf16(x[0]), _to_cl_bf16(x[1]), conv.weight,
+                conv.stride[0], conv.padding[0], 0, want_stats,
+                sh["f"], sh["b"], in_ring, out_ring
+            )
+        elif isinstance(conv, nn.ConvTranspose2d):
+            xl = _to_cl_bf16(x)
             out, stats = ConvT2dNHWCFn.apply(
                 xl, conv.weight, None, conv.stride[0], conv.padding[0],
                 0, want_stats, sh["f"], sh["b"], in_ring, out_ring
             )
         else:
+            xl = _to_cl_bf16(x)
             out, stats = Conv2dNHWCFn.apply(
                 xl, conv.weight, None, conv.stride[0], conv.padding[0],
                 0, want_stats, sh["f"], sh["b"], in_ring, out_ring
@@ -211,16 +221,34 @@ class FusedSequential(nn.Sequential):
         self._pvg_pad_out = pad_out
 
     def forward(self, x):
-        from .conv import _use_hip_path
+        from .conv import _use_hip_path, cat_conv_eligible
 
         mods = list(self)
         i = 0
         while i < len(mods):
             m = mods[i]
+            if isinstance(x, tuple):
+                # dual-source pair from _skip_join: consumed by the fused
+                # conv below, or materialized as a plain concat otherwise
+                fused_dual = (
+                    i + 2 < len(mods)
+                    and cat_conv_eligible(x[0], x[1], m)
+                    and _use_hip_path(x[0])
+                    and isinstance(mods[i + 1], nn.BatchNorm2d)
+                    and act_code(mods[i + 2]) is not None
+                    and mods[i + 1].num_features % 8 == 0
+                    and 256 % (mods[i + 1].num_features // 8) == 0
+                )
+                if not fused_dual:
+                    pa = getattr(x[0], "_pvg_pad", 0)
+                    x = torch.cat(x, 1)
+                    if pa:
+                        x._pvg_pad = pa
             if (
-                isinstance(m, (nn.Conv2d, nn.ConvTranspose2d))
-                and _use_hip_path(x)
-                and _conv_supported(m, x)
+                isinstance(x, tuple)
+                or (isinstance(m, (nn.Conv2d, nn.ConvTranspose2d))
+                    and _use_hip_path(x)
+                    and _conv_supported(m, x))
             ):
                 if (
                     i + 2 < len(mods)

@@ -162,3 +162,49 @@ def test_glds_padded_output_placement(ext):
     ref = F.conv2d(x.float(), w.float(), None, 1, 1)
     err = (got.float()[:, :, 1:-1, 1:-1] - ref).abs().max().item()
     assert err < 0.02 * ref.abs().max().item() + 0.02
+
+
+def test_cat_conv_dual_source_matches(ext):
+    """Dual-pointer glds conv == conv of the materialized concat, forward
+    and both backward operands (dgrad split + dual-X wgrad)."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(3)
+    n, c1, c2, h, k = 4, 64, 128, 16, 128
+    x1 = torch.randn(n, c1, h, h, device="cuda")
+    x2 = torch.randn(n, c2, h, h, device="cuda")
+    w = torch.randn(k, c1 + c2, 3, 3, device="cuda") * 0.05
+    g = torch.randn(n, k, h, h, device="cuda")
+
+    # reference on the dense concat (fp32)
+    xr = torch.cat([x1, x2], 1).bfloat16().float().requires_grad_()
+    wr = w.bfloat16().float().requires_grad_()
+    out_ref = F.conv2d(xr, wr, None, 1, 1)
+    out_ref.backward(g)
+
+    x1p = F.pad(x1, (1,) * 4).bfloat16().contiguous(memory_format=CL)
+    x2p = F.pad(x2, (1,) * 4).bfloat16().contiguous(memory_format=CL)
+    wl = w.bfloat16().contiguous(memory_format=CL)
+    got, _ = ext.conv2d_glds_fwd(x1p, wl, None, 1, 0, False, 0, 0, 0, 0, x2p)
+    err = (got.float() - out_ref).abs().max().item()
+    scale = out_ref.abs().max().item()
+    assert err < 0.02 * scale + 0.02, f"fwd err {err}"
+
+    # dgrad: dual destination (padded outputs, interior compared)
+    gp = F.pad(g, (1,) * 4).bfloat16().contiguous(memory_format=CL)
+    wt = wl.flip(2, 3).transpose(0, 1).contiguous(memory_format=CL)
+    dx2 = torch.empty_like(x2p)
+    dx1, _ = ext.conv2d_glds_fwd(gp, wt, None, 1, 0, False, h + 2, h + 2,
+                                 1, 1, None, dx2)
+    ref_dx = xr.grad
+    e1 = (dx1.float()[:, :, 1:-1, 1:-1] - ref_dx[:, :c1]).abs().max().item()
+    e2 = (dx2.float()[:, :, 1:-1, 1:-1] - ref_dx[:, c1:]).abs().max().item()
+    s = ref_dx.abs().max().item()
+    assert e1 < 0.02 * s + 0.02 and e2 < 0.02 * s + 0.02, (e1, e2)
+
+    # wgrad: dual-X gather (Y ring = 1)
+    ws = ext.conv2d_nhwc_wgrad(gp, x1p, 3, 3, 1, 0, 0, None, 1, x2p)
+    dw = ws.permute(0, 3, 1, 2)
+    e = (dw.float() - wr.grad).abs().max().item()
+    sw = wr.grad.abs().max().item()
+    assert e < 0.02 * sw + 0.02, f"wgrad err {e}"
