@@ -532,8 +532,11 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
   TORCH_CHECK(X.size(0) == Nb, "batch mismatch");
   const long E = (long)B * R * S * A;
 
-  const int BTB = B >= 128 ? 128 : 64;
-  const int BTA = A >= 128 ? 128 : 64;
+  // prefer 128-wide channel tiles; fall back to 64 when the channel count
+  // is not 128-aligned (e.g. dual-X totals like 192) so the glds path's
+  // exact-tiling requirement still holds
+  const int BTB = (B >= 128 && B % 128 == 0) ? 128 : 64;
+  const int BTA = (A >= 128 && A % 128 == 0) ? 128 : 64;
   const int bt = ceil_div(B, BTB), at = ceil_div(A, BTA);
   const int p_total = Nb * HO * WO;
   int sp = (int)splitp;
