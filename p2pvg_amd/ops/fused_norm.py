@@ -229,7 +229,9 @@ class FusedSequential(nn.Sequential):
             m = mods[i]
             if isinstance(x, tuple):
                 # dual-source pair from _skip_join: consumed by the fused
-                # conv below, or materialized as a plain concat otherwise
+                # conv below, passed through to a block whose own inner
+                # FusedSequential will consume it (vgg_layer and friends
+                # forward to .main), or materialized as a plain concat
                 fused_dual = (
                     i + 2 < len(mods)
                     and cat_conv_eligible(x[0], x[1], m)
@@ -239,7 +241,8 @@ class FusedSequential(nn.Sequential):
                     and mods[i + 1].num_features % 8 == 0
                     and 256 % (mods[i + 1].num_features // 8) == 0
                 )
-                if not fused_dual:
+                inner = getattr(m, "main", None)
+                if not fused_dual and not isinstance(inner, FusedSequential):
                     pa = getattr(x[0], "_pvg_pad", 0)
                     x = torch.cat(x, 1)
                     if pa:
