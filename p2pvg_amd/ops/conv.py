@@ -401,10 +401,14 @@ def cat_conv_eligible(x1, x2, conv) -> bool:
     if not isinstance(conv, nn.Conv2d) or isinstance(conv, nn.ConvTranspose2d):
         return False
     C = x1.shape[1] + x2.shape[1]
+    H = x1.shape[2] - 2 * ring
+    W = x1.shape[3] - 2 * ring
+    pow2 = H > 0 and (H & (H - 1)) == 0 and (W & (W - 1)) == 0
     return (conv.kernel_size[0] == 3 and conv.stride[0] == 1
             and conv.padding[0] == ring and x1.shape[1] % 64 == 0
             and x2.shape[1] % 64 == 0 and conv.out_channels % 64 == 0
-            and conv.in_channels == C and x1.shape[2:] == x2.shape[2:])
+            and conv.in_channels == C and x1.shape[2:] == x2.shape[2:]
+            and pow2)   # the dual-X wgrad needs the shift-only pixel decode
 
 
 class ConvT2dNHWCFn(torch.autograd.Function):

@@ -557,7 +557,8 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
 
   const bool pow2 = ((HO * WO) & (HO * WO - 1)) == 0 && (WO & (WO - 1)) == 0;
   const bool use_glds = pad == 0 && pow2 && HO > 0 && WO > 0 &&
-                        B % BTB == 0 && A % BTA == 0 && p_total >= 4 * PCH;
+                        B % BTB == 0 && A % BTA == 0 &&
+                        (X2.has_value() || p_total >= 4 * PCH);
   TORCH_CHECK(!X2.has_value() || use_glds,
               "wgrad: dual-X requires the glds-eligible geometry");
   if (use_glds) {
