@@ -52,9 +52,9 @@ def main():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
-    p.add_argument("--batch", type=int, default=448, help="per-GPU batch size "
-               "(448: best measured throughput with memory headroom for DDP; "
-               "512 fits 1-GPU at 11.9k f/s but reserves the whole 288 GB)")
+    p.add_argument("--batch", type=int, default=512, help="per-GPU batch size "
+               "(512: 13.4k f/s with ~90 GB headroom after the round-2 "
+               "memory work; 640 measured within noise of 512)")
     p.add_argument("--seq_len", type=int, default=30)
     p.add_argument("--g_dim", type=int, default=128)
     p.add_argument("--backbone", type=str, default="vgg")
