@@ -50,9 +50,27 @@ class LSTMStack(nn.Module):
         ]
         return self.hidden
 
-    def _run_cells(self, x: torch.Tensor) -> torch.Tensor:
+    def _fused_heads(self, x) -> bool:
+        t = x[0] if isinstance(x, tuple) else x
+        return (t.is_cuda and ops.backend_mode() != "torch"
+                and ops.hip_available())
+
+    def _embed_in(self, x) -> torch.Tensor:
+        if isinstance(x, tuple):
+            # concat-free embed: the kernel gathers [h | global | t | dt]
+            # from the four sources directly (SURVEY K9+K10)
+            if self._fused_heads(x):
+                from ..ops.lstm_heads import Affine4Fn
+
+                h, g, s1, s2 = x
+                return Affine4Fn.apply(h, g, s1, s2, self.embed.weight,
+                                       self.embed.bias)
+            x = torch.cat(list(x), 1)
+        return self.embed(x.view(-1, self.input_size))
+
+    def _run_cells(self, x) -> torch.Tensor:
         assert self.hidden is not None, "call init_hidden() before forward()"
-        h_in = self.embed(x.view(-1, self.input_size))
+        h_in = self._embed_in(x)
         for i, cell in enumerate(self.lstm):
             self.hidden[i] = ops.lstm_cell(
                 h_in,
@@ -75,8 +93,14 @@ class lstm(LSTMStack):
         self.batch_size = batch_size  # kept for checkpoint/API parity; unused
         self.output = nn.Sequential(nn.Linear(hidden_size, output_size), nn.Tanh())
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.output(self._run_cells(x))
+    def forward(self, x) -> torch.Tensor:
+        h = self._run_cells(x)
+        if self._fused_heads(x):
+            from ..ops.lstm_heads import TanhHeadFn
+
+            return TanhHeadFn.apply(h, self.output[0].weight,
+                                    self.output[0].bias)
+        return self.output(h)
 
 
 class gaussian_lstm(LSTMStack):
@@ -95,8 +119,16 @@ class gaussian_lstm(LSTMStack):
         eps = torch.randn_like(std)
         return eps * std + mu
 
-    def forward(self, x: torch.Tensor):
+    def forward(self, x):
         h = self._run_cells(x)
+        if self._fused_heads(x):
+            from ..ops.lstm_heads import GaussHeadFn
+
+            eps = torch.randn(h.shape[0], self.output_size, device=h.device,
+                              dtype=torch.float32)
+            return GaussHeadFn.apply(h, self.mu_net.weight, self.mu_net.bias,
+                                     self.logvar_net.weight,
+                                     self.logvar_net.bias, eps)
         mu = self.mu_net(h)
         logvar = self.logvar_net(h)
         z = self.reparameterize(mu, logvar)

@@ -236,23 +236,24 @@ class P2PModel(nn.Module):
                 # the predictor output that predicted that frame's latent
                 align_loss = align_loss + self.align_criterion(h, h_pred)
 
-            h_cpaw = torch.cat([h, global_z, time_until_cp, delta_time], 1)
-            h_target_cpaw = torch.cat(
-                [h_target, global_z, time_until_cp, delta_time], 1
-            )
+            # LSTM inputs go in as (h, global, t, dt) TUPLES: the fused
+            # embed kernel gathers the four parts directly (no concat
+            # kernels); the CPU/torch path cats inside the stack
+            h_cpaw = (h, global_z, time_until_cp, delta_time)
+            h_target_cpaw = (h_target, global_z, time_until_cp, delta_time)
 
             zt, mu, logvar = self.posterior(h_target_cpaw)
             zt_p, mu_p, logvar_p = self.prior(h_cpaw)
 
             h_pred = self.frame_predictor(
-                torch.cat([h, zt, time_until_cp, delta_time], 1)
+                (h, zt, time_until_cp, delta_time)
             )
             x_pred = self.decoder([h_pred, skip])
 
             if k == n - 1:
                 # control-point consistency (reference models/p2p_model.py:251-254)
                 h_pred_p = self.frame_predictor(
-                    torch.cat([h, zt_p, time_until_cp, delta_time], 1)
+                    (h, zt_p, time_until_cp, delta_time)
                 )
                 x_pred_p = self.decoder([h_pred_p, skip])
                 cpc_loss = ops.frame_mse(x_pred_p, x_cp)
@@ -471,32 +472,25 @@ class P2PModel(nn.Module):
             else:
                 h = h[0]
 
-            h_cpaw = torch.cat([h, global_z, time_until_cp, delta_time], 1)
+            h_cpaw = (h, global_z, time_until_cp, delta_time)
 
             if i < cfg.n_past:
                 # warm-up: drive the recurrence with ground truth
                 h_target = self.encoder(x[i])[0]
-                h_target_cpaw = torch.cat(
-                    [h_target, global_z, time_until_cp, delta_time], 1
-                )
+                h_target_cpaw = (h_target, global_z, time_until_cp, delta_time)
                 zt, _, _ = self.posterior(h_target_cpaw)
                 zt_p, _, _ = self.prior(h_cpaw)
                 if model_mode in ("posterior", "full"):
-                    self.frame_predictor(
-                        torch.cat([h, zt, time_until_cp, delta_time], 1)
-                    )
+                    self.frame_predictor((h, zt, time_until_cp, delta_time))
                 else:
-                    self.frame_predictor(
-                        torch.cat([h, zt_p, time_until_cp, delta_time], 1)
-                    )
+                    self.frame_predictor((h, zt_p, time_until_cp, delta_time))
                 x_in = x[i]
                 gen_seq.append(x_in)
             else:
                 if i < len(x):
                     h_target = self.encoder(x[i])[0]
-                    h_target_cpaw = torch.cat(
-                        [h_target, global_z, time_until_cp, delta_time], 1
-                    )
+                    h_target_cpaw = (h_target, global_z, time_until_cp,
+                                     delta_time)
                 else:
                     h_target_cpaw = h_cpaw
 
@@ -504,13 +498,9 @@ class P2PModel(nn.Module):
                 zt_p, _, _ = self.prior(h_cpaw)
 
                 if model_mode == "posterior":
-                    h = self.frame_predictor(
-                        torch.cat([h, zt, time_until_cp, delta_time], 1)
-                    )
+                    h = self.frame_predictor((h, zt, time_until_cp, delta_time))
                 else:  # prior | full
-                    h = self.frame_predictor(
-                        torch.cat([h, zt_p, time_until_cp, delta_time], 1)
-                    )
+                    h = self.frame_predictor((h, zt_p, time_until_cp, delta_time))
 
                 x_in = self.decoder([h, skip])
                 gen_seq.append(x_in)

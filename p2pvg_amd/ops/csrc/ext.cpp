@@ -23,6 +23,28 @@ std::vector<torch::Tensor> gaussian_kl_bwd(torch::Tensor mu1, torch::Tensor lv1,
                                            torch::Tensor mu2, torch::Tensor lv2,
                                            torch::Tensor dout, double denom);
 
+torch::Tensor affine4_fwd(torch::Tensor h, torch::Tensor g, torch::Tensor s1,
+                          torch::Tensor s2, torch::Tensor W,
+                          c10::optional<torch::Tensor> b);
+std::vector<torch::Tensor> affine4_bwd(
+    torch::Tensor gout, torch::Tensor h, torch::Tensor g, torch::Tensor s1,
+    torch::Tensor s2, torch::Tensor W, c10::optional<torch::Tensor> dW_acc,
+    c10::optional<torch::Tensor> db_acc, bool need_dh, bool need_dg);
+std::vector<torch::Tensor> gauss_head_fwd(torch::Tensor hin, torch::Tensor Wm,
+                                          torch::Tensor bm, torch::Tensor Wl,
+                                          torch::Tensor bl, torch::Tensor eps);
+torch::Tensor gauss_head_bwd(
+    c10::optional<torch::Tensor> dz, c10::optional<torch::Tensor> dmu,
+    c10::optional<torch::Tensor> dlv, torch::Tensor eps, torch::Tensor lv,
+    torch::Tensor hin, torch::Tensor Wm, torch::Tensor Wl,
+    c10::optional<torch::Tensor> dWm, c10::optional<torch::Tensor> dbm,
+    c10::optional<torch::Tensor> dWl, c10::optional<torch::Tensor> dbl);
+torch::Tensor tanh_head_fwd(torch::Tensor hin, torch::Tensor W,
+                            torch::Tensor b);
+torch::Tensor tanh_head_bwd(torch::Tensor dy, torch::Tensor y,
+                            torch::Tensor hin, torch::Tensor W,
+                            c10::optional<torch::Tensor> dW_acc,
+                            c10::optional<torch::Tensor> db_acc);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 torch::Tensor tr16_probe(long mode);
 std::vector<torch::Tensor> conv2d_nhwc_fwd(torch::Tensor in, torch::Tensor w,
@@ -70,6 +92,13 @@ torch::Tensor sqdiff_sum(torch::Tensor a, torch::Tensor b);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA layout probe");
+  m.def("affine4_fwd", &affine4_fwd, "concat-free 4-source affine (gfx950)");
+  m.def("affine4_bwd", &affine4_bwd, "affine4 backward (managed dW/db)");
+  m.def("gauss_head_fwd", &gauss_head_fwd,
+        "fused mu/logvar/reparameterize head (gfx950)");
+  m.def("gauss_head_bwd", &gauss_head_bwd, "gauss head backward");
+  m.def("tanh_head_fwd", &tanh_head_fwd, "fused Linear+Tanh head (gfx950)");
+  m.def("tanh_head_bwd", &tanh_head_bwd, "tanh head backward");
   m.def("tr16_probe", &tr16_probe, "ds_read_b64_tr_b16 semantics probe");
   m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd,
         "NHWC implicit-GEMM bf16 conv fwd (gfx950 MFMA); returns (out, stats)",

@@ -1,0 +1,121 @@
+"""GPU numerics for the fused LSTM stack heads vs the torch composition."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _cmp(a, b, tol=1e-4, name=""):
+    s = b.abs().max().item() + 1e-6
+    e = (a - b).abs().max().item()
+    assert e < tol * s + 1e-5, f"{name}: err {e} scale {s}"
+
+
+def test_affine4_matches_cat_linear():
+    from p2pvg_amd.ops.lstm_heads import Affine4Fn
+
+    torch.manual_seed(0)
+    B, K1, K2, N = 7, 128, 10, 256
+    h = torch.randn(B, K1, device="cuda", requires_grad=True)
+    g = torch.randn(B, K2, device="cuda", requires_grad=True)
+    s1 = torch.full((B, 1), 0.37, device="cuda")
+    s2 = torch.full((B, 1), 0.11, device="cuda")
+    lin = torch.nn.Linear(K1 + K2 + 2, N).cuda()
+
+    out = Affine4Fn.apply(h, g, s1, s2, lin.weight, lin.bias)
+    gph = torch.randn_like(out)
+    out.backward(gph)
+
+    h2 = h.detach().clone().requires_grad_()
+    g2 = g.detach().clone().requires_grad_()
+    lin2 = torch.nn.Linear(K1 + K2 + 2, N).cuda()
+    lin2.load_state_dict(lin.state_dict())
+    ref = lin2(torch.cat([h2, g2, s1, s2], 1))
+    ref.backward(gph)
+
+    _cmp(out, ref, name="fwd")
+    _cmp(h.grad, h2.grad, name="dh")
+    _cmp(g.grad, g2.grad, name="dg")
+    _cmp(lin.weight.grad, lin2.weight.grad, name="dW")
+    _cmp(lin.bias.grad, lin2.bias.grad, name="db")
+
+
+def test_gauss_head_matches_torch():
+    from p2pvg_amd.ops.lstm_heads import GaussHeadFn
+
+    torch.manual_seed(1)
+    B, K, N = 9, 256, 10
+    hin = torch.randn(B, K, device="cuda", requires_grad=True)
+    mu_net = torch.nn.Linear(K, N).cuda()
+    lv_net = torch.nn.Linear(K, N).cuda()
+    eps = torch.randn(B, N, device="cuda")
+
+    z, mu, lv = GaussHeadFn.apply(hin, mu_net.weight, mu_net.bias,
+                                  lv_net.weight, lv_net.bias, eps)
+    loss = (z * 1.3).sum() + (mu * 0.7).sum() + (lv * -0.4).sum()
+    loss.backward()
+
+    h2 = hin.detach().clone().requires_grad_()
+    mu2 = torch.nn.Linear(K, N).cuda(); mu2.load_state_dict(mu_net.state_dict())
+    lv2 = torch.nn.Linear(K, N).cuda(); lv2.load_state_dict(lv_net.state_dict())
+    m = mu2(h2); l = lv2(h2)
+    zr = eps * (l * 0.5).exp() + m
+    (zr * 1.3).sum().add((m * 0.7).sum()).add((l * -0.4).sum()).backward()
+
+    _cmp(z, zr, name="z")
+    _cmp(mu, m, name="mu")
+    _cmp(lv, l, name="lv")
+    _cmp(hin.grad, h2.grad, name="dh")
+    _cmp(mu_net.weight.grad, mu2.weight.grad, name="dWm")
+    _cmp(lv_net.weight.grad, lv2.weight.grad, name="dWl")
+    _cmp(mu_net.bias.grad, mu2.bias.grad, name="dbm")
+    _cmp(lv_net.bias.grad, lv2.bias.grad, name="dbl")
+
+
+def test_tanh_head_matches_torch():
+    from p2pvg_amd.ops.lstm_heads import TanhHeadFn
+
+    torch.manual_seed(2)
+    B, K, N = 11, 256, 128
+    hin = torch.randn(B, K, device="cuda", requires_grad=True)
+    lin = torch.nn.Linear(K, N).cuda()
+    y = TanhHeadFn.apply(hin, lin.weight, lin.bias)
+    gph = torch.randn_like(y)
+    y.backward(gph)
+
+    h2 = hin.detach().clone().requires_grad_()
+    lin2 = torch.nn.Linear(K, N).cuda(); lin2.load_state_dict(lin.state_dict())
+    ref = torch.tanh(lin2(h2))
+    ref.backward(gph)
+
+    _cmp(y, ref, name="fwd")
+    _cmp(hin.grad, h2.grad, name="dh")
+    _cmp(lin.weight.grad, lin2.weight.grad, name="dW")
+    _cmp(lin.bias.grad, lin2.bias.grad, name="db")
+
+
+def test_lstm_stack_tuple_input_matches_cat():
+    """Whole gaussian stack: tuple (fused) vs materialized cat (torch path),
+    same weights, same eps via seed control at the stack level is not
+    possible (different RNG call shapes) — compare mu/logvar only."""
+    from p2pvg_amd.models.lstm import gaussian_lstm
+
+    torch.manual_seed(3)
+    stack = gaussian_lstm(258, 10, 256, 1, 4).cuda()
+    stack.init_hidden(4, "cuda")
+    h = torch.randn(4, 128, device="cuda")
+    g = torch.randn(4, 128, device="cuda")
+    s1 = torch.full((4, 1), 0.5, device="cuda")
+    s2 = torch.full((4, 1), 0.25, device="cuda")
+    _, mu_a, lv_a = stack((h, g, s1, s2))
+
+    import os
+
+    stack.init_hidden(4, "cuda")
+    os.environ["P2PVG_KERNELS"] = "torch"
+    try:
+        _, mu_b, lv_b = stack(torch.cat([h, g, s1, s2], 1))
+    finally:
+        os.environ.pop("P2PVG_KERNELS", None)
+    _cmp(mu_a, mu_b, tol=5e-3, name="mu")
+    _cmp(lv_a, lv_b, tol=5e-3, name="lv")
