@@ -342,7 +342,7 @@ class P2PModel(nn.Module):
         # pre-step graph, then step. The effective update rule is unchanged:
         # non-prior params move by -lr*Adam(dL1), prior by -lr*Adam(dL2).
 
-        from ..ops.conv import no_weight_grads
+        from ..ops.conv import weight_grad_scope
 
         if mode == "reference":
             # two full-graph traversals, exactly as the reference pays them
@@ -363,16 +363,19 @@ class P2PModel(nn.Module):
                     else:
                         p.grad.copy_(g)
         else:
-            # phase 1: grads of loss1 into everything but the prior
-            torch.autograd.backward(loss1, inputs=nonprior, retain_graph=True)
-            # phase 2: grads of loss2 into the prior only — autograd prunes
-            # the traversal to the kld->prior and cpc->prior paths (one
-            # decoder dgrad chain at the cp step + the prior BPTT chain)
-            # instead of re-walking the whole unrolled graph. The pruned
-            # traversal still reaches the decoder/predictor backward nodes
-            # (their dx is on the path); no_weight_grads suppresses their
-            # discarded-anyway weight-gradient work.
-            with no_weight_grads():
+            # phase 1: grads of loss1 into everything but the prior. The
+            # pruned traversal reaches the prior's backward nodes on the
+            # h->encoder path, so the prior's in-kernel weight accumulation
+            # must be scoped OFF here (the reference zeroes those grads
+            # before phase 2, models/p2p_model.py:266).
+            with weight_grad_scope(prior):
+                torch.autograd.backward(loss1, inputs=nonprior,
+                                        retain_graph=True)
+            # phase 2: grads of loss2 into the prior only — the traversal
+            # still reaches decoder/predictor nodes on the cpc path (their
+            # dx is needed); their weight work is scoped off (discarded by
+            # the update rule anyway).
+            with weight_grad_scope(nonprior):
                 torch.autograd.backward(loss2, inputs=prior)
 
         if grad_sync is not None:

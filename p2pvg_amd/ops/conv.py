@@ -26,33 +26,67 @@ import torch.nn as nn
 
 CL = torch.channels_last
 
-# Phase flag for the two-phase backward: during phase 2 (prior-only update)
-# autograd still traverses the decoder/predictor chain on the cpc path and
-# reports needs_input_grad[weight]=True, but those weight grads are discarded
-# by the update rule (reference zeroes them at the next step start). Skipping
-# them here saves the wasted wgrad kernels AND keeps the managed .grad
-# buffers exactly equal to the phase-1 gradients Adam consumes.
-_WEIGHT_GRADS = True
+# Parameter-scoped weight-grad policy for the two-phase backward. The custom
+# backward Functions accumulate weight grads by SIDE EFFECT whenever their
+# node runs — but autograd's pruned traversal
+# (`torch.autograd.backward(inputs=...)`) still visits nodes of the OTHER
+# optimizer group when they sit on the path (phase 1 reaches the prior's
+# embed on the way to the encoder; phase 2 reaches the decoder on the cpc
+# path), and those groups' weight grads must NOT receive that phase's
+# contributions (the reference discards them: models/p2p_model.py:266 +
+# next-step zero_grad). Each phase therefore declares the param set to SKIP;
+# this also saves the skipped kernels' work entirely.
+_WG_SKIP = None  # Optional[frozenset[int]] of param ids to skip
 
 
-class no_weight_grads:
-    """Context manager: suppress weight/bias/BN-param gradient work in the
-    custom backward Functions (used around the phase-2 backward)."""
+class weight_grad_scope:
+    """Context manager: suppress in-kernel weight/bias/BN-param gradient
+    accumulation for the given parameters."""
+
+    def __init__(self, params):
+        self._ids = frozenset(id(p) for p in params)
 
     def __enter__(self):
-        global _WEIGHT_GRADS
-        self._prev = _WEIGHT_GRADS
-        _WEIGHT_GRADS = False
+        global _WG_SKIP
+        self._prev = _WG_SKIP
+        _WG_SKIP = self._ids
         return self
 
     def __exit__(self, *a):
-        global _WEIGHT_GRADS
-        _WEIGHT_GRADS = self._prev
+        global _WG_SKIP
+        _WG_SKIP = self._prev
         return False
 
 
-def weight_grads_enabled() -> bool:
-    return _WEIGHT_GRADS
+class no_weight_grads:
+    """Suppress ALL weight-grad work (kept for generation/eval paths)."""
+
+    def __enter__(self):
+        global _WG_SKIP
+        self._prev = _WG_SKIP
+        _WG_SKIP = _ALL_PARAMS
+        return self
+
+    def __exit__(self, *a):
+        global _WG_SKIP
+        _WG_SKIP = self._prev
+        return False
+
+
+class _AllParams(frozenset):
+    def __contains__(self, item):  # noqa: D105
+        return True
+
+
+_ALL_PARAMS = _AllParams()
+
+
+def weight_grads_enabled(p=None) -> bool:
+    if _WG_SKIP is None:
+        return True
+    if isinstance(_WG_SKIP, _AllParams):
+        return False
+    return p is None or id(p) not in _WG_SKIP
 
 
 def _acc_target(p):
@@ -264,7 +298,7 @@ class Conv2dNHWCFn(torch.autograd.Function):
         if ctx.act != 0:
             gout = _act_bwd_from_y(gout, y, ctx.act).contiguous(memory_format=CL)
 
-        wge = weight_grads_enabled()
+        wge = weight_grads_enabled(ctx.wref)
         dx = dw = db = None
         if ctx.gemm:
             g2 = gout.reshape(gout.shape[0], gout.shape[1])  # (N, K)
@@ -369,7 +403,7 @@ class CatConv2dFn(torch.autograd.Function):
         gout = gout.contiguous(memory_format=CL)
         if gout.dtype != torch.bfloat16:
             gout = gout.to(torch.bfloat16)
-        wge = weight_grads_enabled()
+        wge = weight_grads_enabled(ctx.wref)
         dx1 = dx2 = dw = None
         if ctx.needs_input_grad[0] or ctx.needs_input_grad[1]:
             wt = w if ctx.shadow_bwd else \
@@ -488,7 +522,7 @@ class ConvT2dNHWCFn(torch.autograd.Function):
         if ctx.act != 0:
             gout = _act_bwd_from_y(gout, y, ctx.act).contiguous(memory_format=CL)
 
-        wge = weight_grads_enabled()
+        wge = weight_grads_enabled(ctx.wref)
         dx = dw = db = None
         if ctx.gemm:
             n = x.shape[0]

@@ -68,7 +68,7 @@ class FusedBNActFn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         r = ctx.ring
-        if not weight_grads_enabled():
+        if not weight_grads_enabled(ctx.gref):
             # phase-2 traversal: dx only, param grads are discarded anyway
             dx, _, _ = ext.bn_act_bwd(
                 x, dy, mean, invstd, gamma32, beta32, scale, ctx.act,

@@ -43,8 +43,7 @@ class Affine4Fn(torch.autograd.Function):
     def backward(ctx, gout):
         h, g, s1, s2, w = ctx.saved_tensors
         gout = gout.contiguous()
-        wge = weight_grads_enabled()
-        need_w = ctx.needs_input_grad[4] and wge
+        need_w = ctx.needs_input_grad[4] and weight_grads_enabled(ctx.wref)
         wg = _acc_target(ctx.wref) if need_w else None
         bg = _acc_target(ctx.bref) if need_w else None
         dw = db = None
@@ -81,8 +80,8 @@ class GaussHeadFn(torch.autograd.Function):
     def backward(ctx, dz, dmu, dlv):
         hin, wm, wl, eps, lv = ctx.saved_tensors
         wm_p, bm_p, wl_p, bl_p = ctx.refs
-        wge = weight_grads_enabled()
-        need_w = wge and (ctx.needs_input_grad[1] or ctx.needs_input_grad[3])
+        need_w = weight_grads_enabled(wm_p) and (
+            ctx.needs_input_grad[1] or ctx.needs_input_grad[3])
         accs = [(_acc_target(p) if need_w else None)
                 for p in (wm_p, bm_p, wl_p, bl_p)]
         rets = [None, None, None, None]
@@ -114,8 +113,7 @@ class TanhHeadFn(torch.autograd.Function):
     def backward(ctx, dy):
         hin, w, y = ctx.saved_tensors
         w_p, b_p = ctx.refs
-        wge = weight_grads_enabled()
-        need_w = wge and ctx.needs_input_grad[1]
+        need_w = weight_grads_enabled(w_p) and ctx.needs_input_grad[1]
         wg = _acc_target(w_p) if need_w else None
         bg = _acc_target(b_p) if need_w else None
         dw = db = None
