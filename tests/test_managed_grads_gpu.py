@@ -131,8 +131,14 @@ def test_managed_grads_match_autograd_accumulation():
         with torch.autocast("cuda", dtype=torch.bfloat16):
             mse, kld, cpc, align = model._compute_losses(prev, cur, tun, dts, plan)
         loss = mse + kld * cfg.beta + align * cfg.weight_align
-        nonprior, _ = model._param_groups()
-        torch.autograd.backward(loss, inputs=nonprior)
+        nonprior, prior = model._param_groups()
+        from p2pvg_amd.ops.conv import weight_grad_scope
+
+        # phase-1 semantics: the prior's side-effecting head backwards sit
+        # on the h->encoder path and must be scoped off, exactly as
+        # _backward_and_step does
+        with weight_grad_scope(prior):
+            torch.autograd.backward(loss, inputs=nonprior)
         torch.cuda.synchronize()
         return {n: (p.grad.clone() if p.grad is not None else None)
                 for n, p in model.named_parameters()}
