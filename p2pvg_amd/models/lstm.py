@@ -51,8 +51,11 @@ class LSTMStack(nn.Module):
         return self.hidden
 
     def _fused_heads(self, x) -> bool:
+        import os
+
         t = x[0] if isinstance(x, tuple) else x
         return (t.is_cuda and ops.backend_mode() != "torch"
+                and os.environ.get("P2PVG_LSTM_HEADS", "1") != "0"
                 and ops.hip_available())
 
     def _embed_in(self, x) -> torch.Tensor:

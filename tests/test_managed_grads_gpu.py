@@ -138,10 +138,20 @@ def test_managed_grads_match_autograd_accumulation():
         # on the h->encoder path and must be scoped off, exactly as
         # _backward_and_step does
         with weight_grad_scope(prior):
-            torch.autograd.backward(loss, inputs=nonprior)
+            torch.autograd.backward(loss, inputs=nonprior,
+                                    retain_graph=True)
+        phase1 = {n: (p.grad.clone() if p.grad is not None else None)
+                  for n, p in model.named_parameters()}
+        # phase 2: prior grads of kld + w_cpc*cpc
+        loss2 = kld + cpc * cfg.weight_cpc
+        with weight_grad_scope(nonprior):
+            torch.autograd.backward(loss2, inputs=prior)
         torch.cuda.synchronize()
-        return {n: (p.grad.clone() if p.grad is not None else None)
-                for n, p in model.named_parameters()}
+        out = dict(phase1)
+        for n, p in model.named_parameters():
+            if n.startswith("prior."):
+                out[n] = p.grad.clone() if p.grad is not None else None
+        return out
 
     managed = run(True)
     plain = run(False)
