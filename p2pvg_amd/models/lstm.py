@@ -54,7 +54,12 @@ class LSTMStack(nn.Module):
         import os
 
         t = x[0] if isinstance(x, tuple) else x
-        return (t.is_cuda and ops.backend_mode() != "torch"
+        # dispatch policy: the fused heads win in the LATENCY-bound regime
+        # (small batch, replay-gap dominated); above ~512 rows the hipBLASLt
+        # GEMM heads are faster (measured: h36m batch 1024 -21% fused,
+        # profiles/MEASUREMENTS.md), so large batches keep the library GEMMs
+        return (t.is_cuda and t.shape[0] <= 512
+                and ops.backend_mode() != "torch"
                 and os.environ.get("P2PVG_LSTM_HEADS", "1") != "0"
                 and ops.hip_available())
 
