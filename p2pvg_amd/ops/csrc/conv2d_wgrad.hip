@@ -487,6 +487,114 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_glds_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Small-A wgrad (the first conv: A = nc = 1 or 3 input channels). The MFMA
+// kernels pay for a 64-wide A tile that is ~95% masked there — and that
+// conv's wgrad was 6.6% of the whole headline step. Here the ENTIRE dW
+// accumulator (B*R*S*A <= 2048 cells) lives in the block's registers
+// (<= 8 cells/thread) and plain VALU rank-1 updates accumulate over an
+// LDS-staged pixel tile. Slab stores + the serial combine keep it
+// deterministic like the MFMA paths.
+
+constexpr int SPIX = 32;
+
+__global__ __launch_bounds__(THREADS) void conv2d_wgrad_smalla_kernel(
+    const __bf16* __restrict__ Y, const __bf16* __restrict__ X,
+    float* __restrict__ ws, int Nb, int HO, int WO, int B, int H, int W,
+    int A, int R, int S, int STRIDE, int PAD, int p_per_slab, int yring) {
+  __shared__ struct __align__(16) {
+    float g[SPIX][64 + 1];
+    float xv[SPIX][64];
+  } lds;
+  const int RSA = R * S * A;
+  const int E = B * RSA;
+  const int tid = threadIdx.x;
+  const int p_begin = blockIdx.z * p_per_slab;
+  const int p_total = Nb * HO * WO;
+  const int p_end = min(p_begin + p_per_slab, p_total);
+  const int HOp = HO + 2 * yring;
+  const int WOp = WO + 2 * yring;
+
+  // this thread's dW cells: cell = c*THREADS + tid
+  float acc[12];
+#pragma unroll
+  for (int c = 0; c < 12; ++c) acc[c] = 0.f;
+  const int ncells = (E + THREADS - 1) / THREADS;
+
+  for (int p0 = p_begin; p0 < p_end; p0 += SPIX) {
+    // stage gout rows (SPIX x B) and the RSA gathers (SPIX x RSA)
+    {
+      const int r = tid >> 3;            // 0..31 pixel
+      const int c0 = (tid & 7) * 8;      // 8 channels per thread
+      const int pix = p0 + r;
+      if (pix < p_end) {
+        long yoff = (long)pix * B;
+        if (yring > 0) {
+          const int n = pix / (HO * WO);
+          const int rem = pix - n * (HO * WO);
+          const int ho = rem / WO;
+          const int wo = rem - ho * WO;
+          yoff = (((long)n * HOp + ho + yring) * WOp + wo + yring) * B;
+        }
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+          lds.g[r][c0 + c] = c0 + c < B ? (float)Y[yoff + c0 + c] : 0.f;
+      } else {
+#pragma unroll
+        for (int c = 0; c < 8; ++c) lds.g[r][c0 + c] = 0.f;
+      }
+    }
+    {
+      // xv: each of the 256 threads stages up to ceil(SPIX*RSA/256) values
+      const int nval = SPIX * RSA;
+      for (int i = tid; i < SPIX * 64; i += THREADS) {
+        const int r = i >> 6;            // pixel
+        const int j = i & 63;            // rsa index
+        float v = 0.f;
+        const int pix = p0 + r;
+        if (j < RSA && pix < p_end) {
+          const int n = pix / (HO * WO);
+          const int rem = pix - n * (HO * WO);
+          const int ho = rem / WO;
+          const int wo = rem - ho * WO;
+          const int rs = j / A;
+          const int a = j - rs * A;
+          const int rr = rs / S;
+          const int ss = rs - rr * S;
+          const int hi = ho * STRIDE - PAD + rr;
+          const int wi = wo * STRIDE - PAD + ss;
+          if (hi >= 0 && hi < H && wi >= 0 && wi < W)
+            v = (float)X[((long)n * H * W + (long)hi * W + wi) * A + a];
+        }
+        lds.xv[r][j] = v;
+        (void)nval;
+      }
+    }
+    __syncthreads();
+    for (int r = 0; r < SPIX; ++r) {
+#pragma unroll
+      for (int c = 0; c < 12; ++c) {
+        if (c >= ncells) break;
+        const int cell = c * THREADS + tid;
+        if (cell < E) {
+          const int b = cell / RSA;
+          const int j = cell - b * RSA;
+          acc[c] = fmaf(lds.g[r][b], lds.xv[r][j], acc[c]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  float* slab = ws + (long)blockIdx.z * E;
+#pragma unroll
+  for (int c = 0; c < 12; ++c) {
+    if (c >= ncells) break;
+    const int cell = c * THREADS + tid;
+    if (cell < E) slab[cell] = acc[c];
+  }
+}
+
 // out[e] (+)= sum over slabs ws[z][e] — serial over z (deterministic).
 __global__ __launch_bounds__(256) void wgrad_combine_kernel(
     const float* __restrict__ ws, float* __restrict__ out, int sp, long E,
@@ -561,6 +669,39 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                         (X2.has_value() || p_total >= 4 * PCH);
   TORCH_CHECK(!X2.has_value() || use_glds,
               "wgrad: dual-X requires the glds-eligible geometry");
+  if (!X2.has_value() && A <= 8 && B <= 64 && R * S * A <= 64 &&
+      (long)B * R * S * A <= 3072) {
+    // tiny-A path: register-resident dW, VALU rank-1 updates
+    const long E = (long)B * R * S * A;
+    int spa = (int)std::min<long>(448, ceil_div(p_total, SPIX));
+    const int pps = ceil_div(ceil_div(p_total, spa), SPIX) * SPIX;
+    spa = ceil_div(p_total, pps);
+    auto wsa = torch::empty({spa, B, (long)R, (long)S, A},
+                            Y.options().dtype(torch::kFloat32));
+    dim3 g(1, 1, spa);
+    hipLaunchKernelGGL(conv2d_wgrad_smalla_kernel, g, dim3(THREADS), 0,
+                       stream, reinterpret_cast<const __bf16*>(Y.data_ptr()),
+                       reinterpret_cast<const __bf16*>(X.data_ptr()),
+                       wsa.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
+                       (int)S, (int)stride, (int)pad, pps, (int)yring);
+    torch::Tensor out;
+    int accumulate = 0;
+    if (acc.has_value()) {
+      out = acc.value();
+      TORCH_CHECK(out.scalar_type() == torch::kFloat32 &&
+                      out.is_non_overlapping_and_dense() && out.numel() == E,
+                  "wgrad acc: need dense fp32 tensor with B*R*S*A elements");
+      accumulate = 1;
+    } else {
+      out = torch::empty({B, (long)R, (long)S, A},
+                         Y.options().dtype(torch::kFloat32));
+    }
+    hipLaunchKernelGGL(wgrad_combine_kernel,
+                       dim3((int)std::min<long>(64, (E + 255) / 256)),
+                       dim3(256), 0, stream, wsa.data_ptr<float>(),
+                       out.data_ptr<float>(), spa, E, accumulate);
+    return out;
+  }
   if (use_glds) {
     // PAD==0 means every X gather is in-bounds (padded operand or genuine
     // valid conv): glds 3-buffer pipeline, shift-only pixel decode
