@@ -515,11 +515,21 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_smalla_kernel(
   const int HOp = HO + 2 * yring;
   const int WOp = WO + 2 * yring;
 
-  // this thread's dW cells: cell = c*THREADS + tid
+  // this thread's dW cells: cell = c*THREADS + tid. Decode ONCE — the
+  // inner loop must be pure LDS-read + fma (a per-iteration cell/RSA
+  // divide serialized the first version). Cells beyond E alias (0,0);
+  // their accumulator is garbage but never stored.
   float acc[12];
-#pragma unroll
-  for (int c = 0; c < 12; ++c) acc[c] = 0.f;
+  int bcell[12], jcell[12];
   const int ncells = (E + THREADS - 1) / THREADS;
+#pragma unroll
+  for (int c = 0; c < 12; ++c) {
+    acc[c] = 0.f;
+    const int cell = c * THREADS + tid;
+    const int b = cell < E ? cell / RSA : 0;
+    bcell[c] = b;
+    jcell[c] = cell < E ? cell - b * RSA : 0;
+  }
 
   for (int p0 = p_begin; p0 < p_end; p0 += SPIX) {
     // stage gout rows (SPIX x B) and the RSA gathers (SPIX x RSA)
@@ -574,15 +584,10 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_smalla_kernel(
     for (int r = 0; r < SPIX; ++r) {
 #pragma unroll
       for (int c = 0; c < 12; ++c) {
-        if (c >= ncells) break;
-        const int cell = c * THREADS + tid;
-        if (cell < E) {
-          const int b = cell / RSA;
-          const int j = cell - b * RSA;
-          acc[c] = fmaf(lds.g[r][b], lds.xv[r][j], acc[c]);
-        }
+        acc[c] = fmaf(lds.g[r][bcell[c]], lds.xv[r][jcell[c]], acc[c]);
       }
     }
+    (void)ncells;
     __syncthreads();
   }
 
@@ -669,8 +674,12 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                         (X2.has_value() || p_total >= 4 * PCH);
   TORCH_CHECK(!X2.has_value() || use_glds,
               "wgrad: dual-X requires the glds-eligible geometry");
-  if (!X2.has_value() && A <= 8 && B <= 64 && R * S * A <= 64 &&
-      (long)B * R * S * A <= 3072) {
+  // DISABLED pending tuning: measured a net headline loss vs the masked
+  // MFMA tile (the staging's per-value pixel decode still dominates);
+  // docs/ROADMAP.md records the attempt. Flip the guard to re-enable.
+  const bool use_smalla = false;
+  if (use_smalla && !X2.has_value() && A <= 8 && B <= 64 &&
+      R * S * A <= 64 && (long)B * R * S * A <= 3072) {
     // tiny-A path: register-resident dW, VALU rank-1 updates
     const long E = (long)B * R * S * A;
     int spa = (int)std::min<long>(448, ceil_div(p_total, SPIX));

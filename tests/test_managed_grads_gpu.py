@@ -172,6 +172,8 @@ def test_managed_grads_match_autograd_accumulation():
     (4, 128, 16, 256, 3, 1, 0),    # glds path, dense Y
     (3, 64, 32, 128, 4, 2, 1),     # k4s2 ringed
     (2, 64, 20, 128, 3, 1, 1),     # non-pow2 spatial -> register path
+    (4, 3, 64, 64, 3, 1, 1),       # tiny-A (first conv)
+    (3, 1, 64, 64, 4, 2, 1),       # dcgan first conv, k4s2 A=1
 ])
 def test_wgrad_padded_matches_reference(n, c, h, k, ks, st, yr):
     """wgrad with padded operands (PAD=0, X in-bounds, Y interior-mapped)
