@@ -501,7 +501,8 @@ constexpr int SPIX = 32;
 __global__ __launch_bounds__(THREADS) void conv2d_wgrad_smalla_kernel(
     const __bf16* __restrict__ Y, const __bf16* __restrict__ X,
     float* __restrict__ ws, int Nb, int HO, int WO, int B, int H, int W,
-    int A, int R, int S, int STRIDE, int PAD, int p_per_slab, int yring) {
+    int A, int R, int S, int STRIDE, int PAD, int p_per_slab, int yring,
+    int howo_sh, int wo_sh) {
   __shared__ struct __align__(16) {
     float g[SPIX][64 + 1];
     float xv[SPIX][64];
@@ -540,10 +541,10 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_smalla_kernel(
       if (pix < p_end) {
         long yoff = (long)pix * B;
         if (yring > 0) {
-          const int n = pix / (HO * WO);
-          const int rem = pix - n * (HO * WO);
-          const int ho = rem / WO;
-          const int wo = rem - ho * WO;
+          const int n = pix >> howo_sh;
+          const int rem = pix - (n << howo_sh);
+          const int ho = rem >> wo_sh;
+          const int wo = rem - (ho << wo_sh);
           yoff = (((long)n * HOp + ho + yring) * WOp + wo + yring) * B;
         }
 #pragma unroll
@@ -563,10 +564,10 @@ __global__ __launch_bounds__(THREADS) void conv2d_wgrad_smalla_kernel(
         float v = 0.f;
         const int pix = p0 + r;
         if (j < RSA && pix < p_end) {
-          const int n = pix / (HO * WO);
-          const int rem = pix - n * (HO * WO);
-          const int ho = rem / WO;
-          const int wo = rem - ho * WO;
+          const int n = pix >> howo_sh;
+          const int rem = pix - (n << howo_sh);
+          const int ho = rem >> wo_sh;
+          const int wo = rem - (ho << wo_sh);
           const int rs = j / A;
           const int a = j - rs * A;
           const int rr = rs / S;
@@ -674,12 +675,15 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
                         (X2.has_value() || p_total >= 4 * PCH);
   TORCH_CHECK(!X2.has_value() || use_glds,
               "wgrad: dual-X requires the glds-eligible geometry");
-  // DISABLED pending tuning: measured a net headline loss vs the masked
-  // MFMA tile (the staging's per-value pixel decode still dominates);
-  // docs/ROADMAP.md records the attempt. Flip the guard to re-enable.
-  const bool use_smalla = false;
-  if (use_smalla && !X2.has_value() && A <= 8 && B <= 64 &&
-      R * S * A <= 64 && (long)B * R * S * A <= 3072) {
+  // DISABLED: three iterations (divide decode, hoisted cells, shift-only
+  // decode) all measured net headline losses vs the masked MFMA tile —
+  // the cost model is wrong somewhere (per-thread scalar LDS read volume
+  // is the prime suspect); docs/ROADMAP.md records the attempts. Flip to
+  // re-enable for tuning.
+  const bool use_smalla = false && !X2.has_value() && A <= 8 && B <= 64 &&
+                          R * S * A <= 64 &&
+                          (long)B * R * S * A <= 3072 && pow2;
+  if (use_smalla) {
     // tiny-A path: register-resident dW, VALU rank-1 updates
     const long E = (long)B * R * S * A;
     int spa = (int)std::min<long>(448, ceil_div(p_total, SPIX));
@@ -688,11 +692,15 @@ torch::Tensor conv2d_nhwc_wgrad(torch::Tensor Y, torch::Tensor X, long R,
     auto wsa = torch::empty({spa, B, (long)R, (long)S, A},
                             Y.options().dtype(torch::kFloat32));
     dim3 g(1, 1, spa);
+    int howo_sh = 0, wo_sh = 0;
+    while ((1 << howo_sh) < HO * WO) ++howo_sh;
+    while ((1 << wo_sh) < WO) ++wo_sh;
     hipLaunchKernelGGL(conv2d_wgrad_smalla_kernel, g, dim3(THREADS), 0,
                        stream, reinterpret_cast<const __bf16*>(Y.data_ptr()),
                        reinterpret_cast<const __bf16*>(X.data_ptr()),
                        wsa.data_ptr<float>(), Nb, HO, WO, B, H, W, A, (int)R,
-                       (int)S, (int)stride, (int)pad, pps, (int)yring);
+                       (int)S, (int)stride, (int)pad, pps, (int)yring,
+                       howo_sh, wo_sh);
     torch::Tensor out;
     int accumulate = 0;
     if (acc.has_value()) {
