@@ -19,62 +19,40 @@ from .image import make_grid, save_gif, save_image, to_uint8_hwc
 STD_SCALE = 3
 
 
+def _ring(frames: torch.Tensor, rgb, pad: int) -> torch.Tensor:
+    """A pad-wide colored ring around each frame; frames (..., 3, h, w)."""
+    ringed = frames.new_empty(frames.shape)
+    for ch in range(3):
+        ringed[..., ch, :, :] = rgb[ch]
+    ringed[..., pad:-pad, pad:-pad] = frames[..., pad:-pad, pad:-pad]
+    return ringed
+
+
+# start frame = orange, control-point frame = red (the parity target is the
+# reference's exact border pixels, misc/visualize.py:13-87)
+_START_RGB = (1.0, 165.0 / 255.0, 0.0)
+_CP_RGB = (1.0, 0.0, 0.0)
+
+
 def add_gt_cp_border(seq: torch.Tensor, seq_len: int, output_len: int, padding: int = 3):
-    """Orange border on the start frame, red on the control-point frame.
-    seq: (t,b,c,h,w)."""
-    t, b, c, h, w = seq.shape
-    if c == 1:
+    """Mark the ground-truth row: orange ring on frame 0, red ring on the
+    control-point frame (and on its repeats past seq_len). seq: (t,b,c,h,w)."""
+    if seq.shape[2] == 1:
         seq = seq.repeat(1, 1, 3, 1, 1)
-    start_ix, end_ix = 0, seq_len - 1
-
-    x_start = seq[start_ix]
-    x_end = seq[end_ix]
-
-    x_start_border = torch.zeros_like(x_start)
-    x_start_border[:, 0] = 1.0
-    x_start_border[:, 1] = 165.0 / 255.0
-    x_start_border[:, :, padding : w - padding, padding : w - padding] = x_start[
-        :, :, padding : w - padding, padding : w - padding
-    ]
-
-    x_end_border = torch.zeros_like(x_end)
-    x_end_border[:, 0] = 1.0
-    x_end_border[:, :, padding : w - padding, padding : w - padding] = x_end[
-        :, :, padding : w - padding, padding : w - padding
-    ]
-
-    seq[start_ix] = x_start_border
-    seq[end_ix] = x_end_border
-    for i in range(seq_len, output_len):
-        seq[i] = x_end_border
+    cp = _ring(seq[seq_len - 1], _CP_RGB, padding)
+    seq[0] = _ring(seq[0], _START_RGB, padding)
+    for i in range(seq_len - 1, output_len):
+        seq[i] = cp
     return seq
 
 
 def add_samples_cp_border(samples: torch.Tensor, seq_len: int, output_len: int, padding: int = 3):
-    """samples: (nsample,t,b,c,h,w); borders on frame 0 and frame output_len-1."""
-    ns, t, b, c, h, w = samples.shape
-    if c == 1:
+    """Mark each sample row: orange ring on frame 0, red ring on the LAST
+    generated frame. samples: (nsample,t,b,c,h,w)."""
+    if samples.shape[3] == 1:
         samples = samples.repeat(1, 1, 1, 3, 1, 1)
-    start_ix, end_ix = 0, output_len - 1
-
-    x_start = samples[:, start_ix]
-    x_end = samples[:, end_ix]
-
-    x_start_border = torch.zeros_like(x_start)
-    x_start_border[:, :, 0] = 1.0
-    x_start_border[:, :, 1] = 165.0 / 255.0
-    x_start_border[:, :, :, padding : w - padding, padding : w - padding] = x_start[
-        :, :, :, padding : w - padding, padding : w - padding
-    ]
-
-    x_end_border = torch.zeros_like(x_end)
-    x_end_border[:, :, 0] = 1.0
-    x_end_border[:, :, :, padding : w - padding, padding : w - padding] = x_end[
-        :, :, :, padding : w - padding, padding : w - padding
-    ]
-
-    samples[:, start_ix] = x_start_border
-    samples[:, end_ix] = x_end_border
+    samples[:, 0] = _ring(samples[:, 0], _START_RGB, padding)
+    samples[:, output_len - 1] = _ring(samples[:, output_len - 1], _CP_RGB, padding)
     return samples
 
 
