@@ -121,6 +121,34 @@ class gaussian_lstm(LSTMStack):
         self.batch_size = batch_size
         self.mu_net = nn.Linear(hidden_size, output_size)
         self.logvar_net = nn.Linear(hidden_size, output_size)
+        # per-step stacked-weight cache [Wm; Wl] for the fused head: built
+        # lazily, refreshed once per init_hidden (i.e. once per training
+        # step / generation sequence) instead of re-catting per timestep
+        self._stk: Optional[Tuple[torch.Tensor, torch.Tensor]] = None
+        self._stk_dirty = True
+
+    def init_hidden(self, batch_size: int = 1, device=None, dtype=None):
+        self._stk_dirty = True
+        return super().init_hidden(batch_size, device, dtype)
+
+    def _stacked(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        n = self.output_size
+        w = self.mu_net.weight
+        if self._stk is None or self._stk[0].device != w.device:
+            self._stk = (
+                torch.empty(2 * n, w.shape[1], device=w.device, dtype=torch.float32),
+                torch.empty(2 * n, device=w.device, dtype=torch.float32),
+            )
+            self._stk_dirty = True
+        if self._stk_dirty:
+            ws, bs = self._stk
+            with torch.no_grad():  # grads route via the managed accumulation
+                ws[:n].copy_(self.mu_net.weight)
+                ws[n:].copy_(self.logvar_net.weight)
+                bs[:n].copy_(self.mu_net.bias)
+                bs[n:].copy_(self.logvar_net.bias)
+            self._stk_dirty = False
+        return self._stk
 
     def reparameterize(self, mu: torch.Tensor, logvar: torch.Tensor) -> torch.Tensor:
         std = logvar.mul(0.5).exp()
@@ -134,9 +162,10 @@ class gaussian_lstm(LSTMStack):
 
             eps = torch.randn(h.shape[0], self.output_size, device=h.device,
                               dtype=torch.float32)
+            ws, bs = self._stacked()
             return GaussHeadFn.apply(h, self.mu_net.weight, self.mu_net.bias,
                                      self.logvar_net.weight,
-                                     self.logvar_net.bias, eps)
+                                     self.logvar_net.bias, eps, ws, bs)
         mu = self.mu_net(h)
         logvar = self.logvar_net(h)
         z = self.reparameterize(mu, logvar)

@@ -32,11 +32,13 @@ std::vector<torch::Tensor> affine4_bwd(
     c10::optional<torch::Tensor> db_acc, bool need_dh, bool need_dg);
 std::vector<torch::Tensor> gauss_head_fwd(torch::Tensor hin, torch::Tensor Wm,
                                           torch::Tensor bm, torch::Tensor Wl,
-                                          torch::Tensor bl, torch::Tensor eps);
+                                          torch::Tensor bl, torch::Tensor eps,
+                                          c10::optional<torch::Tensor> Ws_c,
+                                          c10::optional<torch::Tensor> bs_c);
 torch::Tensor gauss_head_bwd(
     c10::optional<torch::Tensor> dz, c10::optional<torch::Tensor> dmu,
     c10::optional<torch::Tensor> dlv, torch::Tensor eps, torch::Tensor lv,
-    torch::Tensor hin, torch::Tensor Wm, torch::Tensor Wl,
+    torch::Tensor hin, torch::Tensor Ws, long N_,
     c10::optional<torch::Tensor> dWm, c10::optional<torch::Tensor> dbm,
     c10::optional<torch::Tensor> dWl, c10::optional<torch::Tensor> dbl);
 torch::Tensor tanh_head_fwd(torch::Tensor hin, torch::Tensor W,

@@ -212,7 +212,7 @@ __global__ __launch_bounds__(THREADS) void head_dw_kernel(
     const float* __restrict__ gout, const float* __restrict__ h,
     const float* __restrict__ g, const float* __restrict__ s1,
     const float* __restrict__ s2, float* __restrict__ dW,
-    float* __restrict__ db, int B, int N, int K1, int K2, int Kw,
+    float* __restrict__ db, int B, int N, int ldg, int K1, int K2, int Kw,
     int accumulate) {
   __shared__ struct __align__(16) {
     float gt[RB][TC + 1];   // gout tile (r, n)
@@ -238,7 +238,7 @@ __global__ __launch_bounds__(THREADS) void head_dw_kernel(
       const int r = tid >> 3;
       const int c0 = (tid & 7) * 8;
       if (r < rows) {
-        const float* gr = gout + (long)(r0 + r) * N + n0;
+        const float* gr = gout + (long)(r0 + r) * ldg + n0;
 #pragma unroll
         for (int c = 0; c < 8; ++c)
           lds.gt[r][c0 + c] = n0 + c0 + c < N ? gr[c0 + c] : 0.f;
@@ -295,17 +295,22 @@ __global__ __launch_bounds__(THREADS) void head_dw_kernel(
 }
 
 // pointwise: dmu_t = dmu + dz, dlv_t = dlv + dz*eps*exp(lv/2)/2
+// writes the two halves of the STACKED (B, 2N) grad directly (row stride
+// 2N): no temporaries, no narrow-copy passes
 __global__ __launch_bounds__(THREADS) void gauss_combine_kernel(
     const float* __restrict__ dz, const float* __restrict__ dmu,
     const float* __restrict__ dlv, const float* __restrict__ eps,
-    const float* __restrict__ lv, float* __restrict__ dmu_t,
-    float* __restrict__ dlv_t, long nel) {
+    const float* __restrict__ lv, float* __restrict__ dst, int N,
+    long nel) {
   for (long i = (long)blockIdx.x * THREADS + threadIdx.x; i < nel;
        i += (long)gridDim.x * THREADS) {
+    const long r = i / N;
+    const int n = (int)(i - r * N);
     const float dzv = dz != nullptr ? dz[i] : 0.f;
-    dmu_t[i] = (dmu != nullptr ? dmu[i] : 0.f) + dzv;
-    dlv_t[i] = (dlv != nullptr ? dlv[i] : 0.f) +
-               dzv * eps[i] * 0.5f * __expf(0.5f * lv[i]);
+    dst[r * 2 * N + n] = (dmu != nullptr ? dmu[i] : 0.f) + dzv;
+    dst[r * 2 * N + N + n] =
+        (dlv != nullptr ? dlv[i] : 0.f) +
+        dzv * eps[i] * 0.5f * __expf(0.5f * lv[i]);
   }
 }
 
@@ -378,21 +383,25 @@ std::vector<torch::Tensor> affine4_bwd(
                        s1.data_ptr<float>(), s2.data_ptr<float>(),
                        dW_acc->data_ptr<float>(),
                        db_acc.has_value() ? db_acc->data_ptr<float>() : nullptr,
-                       B, N, K1, K2, Kw, 1);
+                       B, N, N, K1, K2, Kw, 1);
   }
   return {dh, dg};
 }
 
 std::vector<torch::Tensor> gauss_head_fwd(torch::Tensor hin, torch::Tensor Wm,
                                           torch::Tensor bm, torch::Tensor Wl,
-                                          torch::Tensor bl,
-                                          torch::Tensor eps) {
+                                          torch::Tensor bl, torch::Tensor eps,
+                                          c10::optional<torch::Tensor> Ws_c,
+                                          c10::optional<torch::Tensor> bs_c) {
   chk(hin); chk(Wm); chk(Wl);
   const int B = hin.size(0), K = hin.size(1), N = Wm.size(0);
   TORCH_CHECK(2 * N <= TC, "gauss head: 2*z_dim must fit one column tile");
-  // stacked weights/biases: rows [0,N) = mu head, [N,2N) = lv head
-  auto Ws = torch::cat({Wm, Wl}, 0).contiguous();
-  auto bs = torch::cat({bm, bl}, 0).contiguous();
+  // stacked weights/biases: rows [0,N) = mu head, [N,2N) = lv head —
+  // either the caller's per-step cache or a fresh cat
+  auto Ws = Ws_c.has_value() ? Ws_c.value()
+                             : torch::cat({Wm, Wl}, 0).contiguous();
+  auto bs = bs_c.has_value() ? bs_c.value()
+                             : torch::cat({bm, bl}, 0).contiguous();
   auto mu = torch::empty({B, N}, hin.options());
   auto lv = torch::empty({B, N}, hin.options());
   auto z = torch::empty({B, N}, hin.options());
@@ -403,65 +412,53 @@ std::vector<torch::Tensor> gauss_head_fwd(torch::Tensor hin, torch::Tensor Wm,
                      eps.data_ptr<float>(), mu.data_ptr<float>(),
                      lv.data_ptr<float>(), z.data_ptr<float>(), B, 2 * N, K,
                      0, 2);
-  return {mu, lv, z};
+  return {mu, lv, z, Ws};
 }
 
 // returns dh; accumulates the four head-param grads when buffers present
 torch::Tensor gauss_head_bwd(
     c10::optional<torch::Tensor> dz, c10::optional<torch::Tensor> dmu,
     c10::optional<torch::Tensor> dlv, torch::Tensor eps, torch::Tensor lv,
-    torch::Tensor hin, torch::Tensor Wm, torch::Tensor Wl,
+    torch::Tensor hin, torch::Tensor Ws, long N_,
     c10::optional<torch::Tensor> dWm, c10::optional<torch::Tensor> dbm,
     c10::optional<torch::Tensor> dWl, c10::optional<torch::Tensor> dbl) {
-  const int B = hin.size(0), K = hin.size(1), N = Wm.size(0);
+  const int B = hin.size(0), K = hin.size(1), N = (int)N_;
   auto stream = at::cuda::getCurrentCUDAStream();
-  // stacked head-grad (B, 2N): [dmu_t | dlv_t]
+  // stacked head-grad (B, 2N): [dmu_t | dlv_t], written strided in one pass
   auto dst = torch::empty({B, 2 * N}, hin.options());
-  auto dmu_t = dst.narrow(1, 0, N);
-  auto dlv_t = dst.narrow(1, N, N);
   const long nel = (long)B * N;
-  {
-    // write the two halves via strided narrow views is non-contiguous;
-    // run the combine into temporaries then pack
-    auto tmp_mu = torch::empty({B, N}, hin.options());
-    auto tmp_lv = torch::empty({B, N}, hin.options());
-    hipLaunchKernelGGL(gauss_combine_kernel,
-                       dim3((int)std::min<long>(2048, (nel + THREADS - 1) / THREADS)),
-                       dim3(THREADS), 0, stream,
-                       dz.has_value() ? dz->data_ptr<float>() : nullptr,
-                       dmu.has_value() ? dmu->data_ptr<float>() : nullptr,
-                       dlv.has_value() ? dlv->data_ptr<float>() : nullptr,
-                       eps.data_ptr<float>(), lv.data_ptr<float>(),
-                       tmp_mu.data_ptr<float>(), tmp_lv.data_ptr<float>(),
-                       nel);
-    dmu_t.copy_(tmp_mu);
-    dlv_t.copy_(tmp_lv);
-  }
-  auto Ws = torch::cat({Wm, Wl}, 0).contiguous();
+  hipLaunchKernelGGL(gauss_combine_kernel,
+                     dim3((int)std::min<long>(2048, (nel + THREADS - 1) / THREADS)),
+                     dim3(THREADS), 0, stream,
+                     dz.has_value() ? dz->data_ptr<float>() : nullptr,
+                     dmu.has_value() ? dmu->data_ptr<float>() : nullptr,
+                     dlv.has_value() ? dlv->data_ptr<float>() : nullptr,
+                     eps.data_ptr<float>(), lv.data_ptr<float>(),
+                     dst.data_ptr<float>(), N, nel);
   auto dh = torch::empty_like(hin);
   hipLaunchKernelGGL(head_dsrc_kernel, dim3(ceil_div(K, TC), ceil_div(B, RB)),
                      dim3(THREADS), 0, stream, dst.data_ptr<float>(),
                      Ws.data_ptr<float>(), dh.data_ptr<float>(), nullptr, B,
                      2 * N, K, 0, K);
+  // the two dW kernels read their half of the stacked grad IN PLACE
+  // (row stride 2N) — no contiguous() copies
   if (dWm.has_value()) {
     hipLaunchKernelGGL(head_dw_kernel,
                        dim3(ceil_div(N, TC), ceil_div(K + 1, TC)),
-                       dim3(THREADS), 0, stream,
-                       dmu_t.contiguous().data_ptr<float>(),
+                       dim3(THREADS), 0, stream, dst.data_ptr<float>(),
                        hin.data_ptr<float>(), nullptr, nullptr, nullptr,
                        dWm->data_ptr<float>(),
                        dbm.has_value() ? dbm->data_ptr<float>() : nullptr,
-                       B, N, K, 0, K, 1);
+                       B, N, 2 * N, K, 0, K, 1);
   }
   if (dWl.has_value()) {
     hipLaunchKernelGGL(head_dw_kernel,
                        dim3(ceil_div(N, TC), ceil_div(K + 1, TC)),
-                       dim3(THREADS), 0, stream,
-                       dlv_t.contiguous().data_ptr<float>(),
+                       dim3(THREADS), 0, stream, dst.data_ptr<float>() + N,
                        hin.data_ptr<float>(), nullptr, nullptr, nullptr,
                        dWl->data_ptr<float>(),
                        dbl.has_value() ? dbl->data_ptr<float>() : nullptr,
-                       B, N, K, 0, K, 1);
+                       B, N, 2 * N, K, 0, K, 1);
   }
   return dh;
 }
@@ -505,7 +502,7 @@ torch::Tensor tanh_head_bwd(torch::Tensor dy, torch::Tensor y,
                        hin.data_ptr<float>(), nullptr, nullptr, nullptr,
                        dW_acc->data_ptr<float>(),
                        db_acc.has_value() ? db_acc->data_ptr<float>() : nullptr,
-                       B, N, K, 0, K, 1);
+                       B, N, N, K, 0, K, 1);
   }
   return dh;
 }

@@ -63,22 +63,27 @@ class Affine4Fn(torch.autograd.Function):
 
 
 class GaussHeadFn(torch.autograd.Function):
+    """ws/bs: optional per-step stacked-weight cache (rows [Wm; Wl]) owned by
+    the module and refreshed once per step — when given, the forward skips
+    its cat and the backward reuses the SAME stacked tensor (saved in ctx),
+    so no weight re-stacking happens anywhere on the timestep loop."""
+
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.float32)
-    def forward(ctx, hin, wm, bm, wl, bl, eps):
+    def forward(ctx, hin, wm, bm, wl, bl, eps, ws=None, bs=None):
         hin = hin.contiguous()
-        mu, lv, z = _ext().gauss_head_fwd(
+        mu, lv, z, ws_used = _ext().gauss_head_fwd(
             hin, wm.contiguous(), bm.contiguous(), wl.contiguous(),
-            bl.contiguous(), eps.contiguous(),
+            bl.contiguous(), eps.contiguous(), ws, bs,
         )
-        ctx.save_for_backward(hin, wm, wl, eps, lv)
+        ctx.save_for_backward(hin, ws_used, eps, lv)
         ctx.refs = (wm, bm, wl, bl)
         return z, mu, lv
 
     @staticmethod
     @torch.amp.custom_bwd(device_type="cuda")
     def backward(ctx, dz, dmu, dlv):
-        hin, wm, wl, eps, lv = ctx.saved_tensors
+        hin, ws, eps, lv = ctx.saved_tensors
         wm_p, bm_p, wl_p, bl_p = ctx.refs
         need_w = weight_grads_enabled(wm_p) and (
             ctx.needs_input_grad[1] or ctx.needs_input_grad[3])
@@ -92,10 +97,10 @@ class GaussHeadFn(torch.autograd.Function):
             dz.contiguous() if dz is not None else None,
             dmu.contiguous() if dmu is not None else None,
             dlv.contiguous() if dlv is not None else None,
-            eps, lv, hin, wm, wl, *accs,
+            eps, lv, hin, ws, wm_p.shape[0], *accs,
         )
         return (dh if ctx.needs_input_grad[0] else None,
-                rets[0], rets[1], rets[2], rets[3], None)
+                rets[0], rets[1], rets[2], rets[3], None, None, None)
 
 
 class TanhHeadFn(torch.autograd.Function):
