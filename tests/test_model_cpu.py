@@ -181,3 +181,31 @@ def test_zero_grads_stable_buffers(tiny_cfg):
     for n, p in model.named_parameters():
         assert id(p.grad) == ids[n], f"{n}: grad buffer identity changed"
         assert float(p.grad.abs().max()) == 0.0
+
+
+def test_gauss_stacked_cache_refreshes_on_init_hidden():
+    """The fused-head weight cache [Wm; Wl] must follow the params: stale
+    until init_hidden marks it dirty (one refresh per step/sequence)."""
+    import torch
+
+    from p2pvg_amd.models.lstm import gaussian_lstm
+
+    stack = gaussian_lstm(258, 10, 256, 1, 4)
+    stack.init_hidden(4)
+    ws, bs = stack._stacked()
+    assert ws.shape == (20, 256) and bs.shape == (20,)
+    assert torch.equal(ws[:10], stack.mu_net.weight)
+    assert torch.equal(ws[10:], stack.logvar_net.weight)
+
+    with torch.no_grad():
+        stack.mu_net.weight.add_(1.0)
+    # same step: cache intentionally NOT refreshed (weights only move at
+    # optimizer.step(), which is always followed by a new init_hidden)
+    ws2, _ = stack._stacked()
+    assert ws2 is ws and not torch.equal(ws[:10], stack.mu_net.weight)
+
+    stack.init_hidden(4)  # new step/sequence -> refresh
+    ws3, bs3 = stack._stacked()
+    assert ws3 is ws  # persistent buffer, stable storage for hipGraph replay
+    assert torch.equal(ws[:10], stack.mu_net.weight)
+    assert torch.equal(bs3[:10], stack.mu_net.bias)
