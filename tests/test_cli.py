@@ -142,3 +142,38 @@ def test_read_video_gif_roundtrip(tmp_path):
     t = gen_cli.read_video(str(p))
     assert t.shape[0] == 3 and t.shape[1] == 1 and t.shape[2] == 3
     assert 0.0 <= float(t.min()) and float(t.max()) <= 1.0
+
+
+def test_train_resume_from_checkpoint(tmp_path):
+    """train.py --ckpt resumes into the SAME log dir at the next epoch
+    (reference train.py resume contract: log dir and epoch both come from
+    the checkpoint)."""
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    log_dir = tmp_path / "logs" / "run"
+    flags = ["--dataset", "mnist", "--backbone", "dcgan", "--batch_size", "2",
+             "--max_seq_len", "6", "--delta_len", "1", "--g_dim", "32",
+             "--z_dim", "4", "--rnn_size", "32", "--epoch_size", "2",
+             "--nsample", "2", "--device", "cpu", "--qual_iter", "100",
+             "--data_root", "/nonexistent", "--num_workers", "0",
+             "--log_dir", str(log_dir)]
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "train.py"), "--nepochs", "1",
+         *flags],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-3000:]
+    run = [d for d in (tmp_path / "logs").iterdir() if d.is_dir()][0]
+    assert (run / "model_0.pth").exists()
+    assert not (run / "model_1.pth").exists()
+
+    r2 = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "train.py"), "--nepochs", "2",
+         "--ckpt", str(run / "model.pth"), *flags],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r2.returncode == 0, r2.stderr[-3000:]
+    assert "continuing at epoch 1" in (r2.stdout + r2.stderr)
+    assert (run / "model_1.pth").exists(), "resume did not write epoch-1 ckpt"
+    # resumed run landed in the original log dir, no second dir created
+    assert len([d for d in (tmp_path / "logs").iterdir() if d.is_dir()]) == 1
