@@ -177,3 +177,38 @@ def test_train_resume_from_checkpoint(tmp_path):
     assert (run / "model_1.pth").exists(), "resume did not write epoch-1 ckpt"
     # resumed run landed in the original log dir, no second dir created
     assert len([d for d in (tmp_path / "logs").iterdir() if d.is_dir()]) == 1
+
+
+def test_train_deterministic_cpu_bitwise(tmp_path):
+    """Two identical --deterministic --seed runs produce bitwise-identical
+    weights (the GPU counterpart over the native kernels:
+    tests/test_managed_grads_gpu.py)."""
+    import torch
+
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    outs = []
+    for sub in ("a", "b"):
+        log_dir = tmp_path / sub / "run"
+        r = subprocess.run(
+            [sys.executable, os.path.join(ROOT, "train.py"),
+             "--dataset", "mnist", "--backbone", "dcgan", "--batch_size", "2",
+             "--max_seq_len", "6", "--delta_len", "1", "--g_dim", "32",
+             "--z_dim", "4", "--rnn_size", "32", "--nepochs", "1",
+             "--epoch_size", "2", "--nsample", "2", "--device", "cpu",
+             "--qual_iter", "100", "--deterministic", "--seed", "11",
+             "--data_root", "/nonexistent", "--num_workers", "0",
+             "--log_dir", str(log_dir)],
+            cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+        )
+        assert r.returncode == 0, r.stderr[-3000:]
+        run = [d for d in (tmp_path / sub).iterdir() if d.is_dir()][0]
+        outs.append(torch.load(run / "model.pth", map_location="cpu",
+                               weights_only=False))
+    sd_a = outs[0]["frame_predictor"], outs[0]["encoder"]
+    sd_b = outs[1]["frame_predictor"], outs[1]["encoder"]
+    for a, b in zip(sd_a, sd_b):
+        for k in a:
+            va, vb = a[k], b[k]
+            if isinstance(va, torch.Tensor) and va.is_floating_point():
+                assert torch.equal(va, vb), f"weights diverged at {k}"
