@@ -38,12 +38,20 @@ def get_logger(logpath: str, name: str = "p2pvg_amd", displaying=True, saving=Tr
 
 
 def store_cmd(log_dir: str) -> str:
-    """Write the reproduction command to <log_dir>/cmd.txt
-    (reference misc/utils.py:238-252)."""
+    """Write the reproduction command to <log_dir>/cmd.txt and snapshot the
+    invoking script's source next to it (reference misc/utils.py:227-252
+    stores both the command line and the training source for provenance)."""
     cmd = "python " + " ".join(sys.argv)
     os.makedirs(log_dir, exist_ok=True)
     with open(os.path.join(log_dir, "cmd.txt"), "w") as f:
         f.write(cmd + "\n")
+    script = sys.argv[0] if sys.argv else ""
+    if script and os.path.isfile(script):
+        import shutil
+
+        dst = os.path.join(log_dir, "src")
+        os.makedirs(dst, exist_ok=True)
+        shutil.copy2(script, os.path.join(dst, os.path.basename(script)))
     return cmd
 
 

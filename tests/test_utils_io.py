@@ -54,6 +54,25 @@ def test_store_cmd_writes_argv(tmp_path):
     assert found, "store_cmd wrote nothing"
 
 
+def test_store_cmd_snapshots_entry_script(tmp_path):
+    """store_cmd copies the invoking script's source into <log_dir>/src
+    (reference misc/utils.py:227-229 provenance dump)."""
+    import sys
+
+    from p2pvg_amd.utils import store_cmd
+
+    script = tmp_path / "fake_train.py"
+    script.write_text("print('hi')\n")
+    argv0 = sys.argv[0]
+    sys.argv[0] = str(script)
+    try:
+        store_cmd(str(tmp_path / "log"))
+    finally:
+        sys.argv[0] = argv0
+    snap = tmp_path / "log" / "src" / "fake_train.py"
+    assert snap.exists() and snap.read_text() == "print('hi')\n"
+
+
 def test_get_logger_writes_file(tmp_path):
     from p2pvg_amd.utils import get_logger
 
