@@ -212,3 +212,28 @@ def test_train_deterministic_cpu_bitwise(tmp_path):
             va, vb = a[k], b[k]
             if isinstance(va, torch.Tensor) and va.is_floating_point():
                 assert torch.equal(va, vb), f"weights diverged at {k}"
+
+
+def test_bench_distributed_2rank_gloo_cpu():
+    """The exact multi-rank path the driver launches for SCALE runs:
+    torch.distributed.run -> bench.py with world=2 (gloo on CPU here, RCCL
+    on GPU boxes) — init, DDPGradSync, barriers, MAX-over-ranks timing,
+    one JSON line from rank 0 with parallelism dp2."""
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29771", os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1", "--batch", "2",
+         "--seq_len", "6"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=900,
+    )
+    assert r.returncode == 0, r.stderr[-3000:]
+    lines = [ln for ln in r.stdout.strip().splitlines()
+             if ln.startswith("{")]
+    assert len(lines) == 1, f"expected exactly one JSON line: {r.stdout}"
+    d = json.loads(lines[0])
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 4  # 2 ranks x per-rank batch 2
+    assert d["value"] > 0 and d["ms_per_step"] > 0
