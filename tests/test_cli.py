@@ -237,3 +237,29 @@ def test_bench_distributed_2rank_gloo_cpu():
     assert d["config"]["parallelism"] == "dp2"
     assert d["config"]["global_batch"] == 4  # 2 ranks x per-rank batch 2
     assert d["value"] > 0 and d["ms_per_step"] > 0
+
+
+def test_train_ddp_2rank_gloo_cpu(tmp_path):
+    """train.py under torch.distributed.run with 2 CPU ranks (gloo): real
+    setup_distributed + per-rank data seeds + DDPGradSync in the epoch loop;
+    rank 0 owns logging/checkpoints."""
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    log_dir = tmp_path / "logs" / "run"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29772", os.path.join(ROOT, "train.py"),
+         "--dataset", "mnist", "--backbone", "dcgan", "--batch_size", "2",
+         "--max_seq_len", "6", "--delta_len", "1", "--g_dim", "32",
+         "--z_dim", "4", "--rnn_size", "32", "--nepochs", "1",
+         "--epoch_size", "2", "--nsample", "2", "--device", "cpu",
+         "--qual_iter", "100", "--ddp", "--data_root", "/nonexistent",
+         "--num_workers", "0", "--log_dir", str(log_dir)],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=900,
+    )
+    assert r.returncode == 0, r.stderr[-3000:]
+    run = [d for d in (tmp_path / "logs").iterdir() if d.is_dir()][0]
+    assert (run / "model_0.pth").exists()
+    # exactly one rank wrote the checkpoint/log set
+    assert (run / "cmd.txt").exists() and (run / "scalars.jsonl").exists()
