@@ -185,9 +185,19 @@ def main():
             acc += torch.stack(list(losses))
             frames += (len(x) if not isinstance(x, tuple) else len(x[1])) * cfg.batch_size
 
+            if world > 1 and i % 50 == 0 and i != 0:
+                # average the 4 loss scalars across ranks for logging
+                # (collective: must run on every rank) — SURVEY §5.8
+                import torch.distributed as dist
+
+                lg = acc.clone()
+                dist.all_reduce(lg)
+                lg /= world
+            else:
+                lg = acc
             if is_main and i % 50 == 0 and i != 0:
                 step = epoch * cfg.epoch_size + i
-                vals = (acc / (i + 1)).cpu()
+                vals = (lg / (i + 1)).cpu()
                 for tag, v in zip(("mse", "kld", "cpc", "align"), vals):
                     writer.add_scalar(f"Train/{tag}", v, step)
                 if cfg.log_histograms:
