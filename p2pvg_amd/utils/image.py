@@ -56,9 +56,11 @@ def save_image(tensor: torch.Tensor, fname: str) -> None:
 
 
 def save_gif(fname: str, frames: Sequence[np.ndarray], duration: float = 0.25) -> None:
-    """frames: list of (H,W,C) uint8 arrays."""
+    """frames: list of (H,W,C) uint8 arrays; C=1 saved as grayscale."""
     from PIL import Image
 
+    frames = [f[:, :, 0] if f.ndim == 3 and f.shape[2] == 1 else f
+              for f in frames]
     imgs: List = [Image.fromarray(f) for f in frames]
     imgs[0].save(
         fname,

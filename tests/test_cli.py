@@ -63,6 +63,21 @@ def test_train_then_generate(tmp_path):
         assert (out_dir / f"len_{length}-gt.png").exists()
         assert (out_dir / f"len_{length}-gen_full.gif").exists()
 
+    # --multi_cp and --loop modes (reference README demo flows)
+    for flag, stem in (("--multi_cp", "multicpgen"), ("--loop", "loopgen")):
+        mode_dir = tmp_path / f"gen_out_{stem}"
+        r3 = subprocess.run(
+            [sys.executable, os.path.join(ROOT, "generate.py"),
+             "--ckpt", str(ckpt), "--video", str(frames_dir),
+             "--output_root", str(mode_dir), "--device", "cpu", flag,
+             "--segment_len", "4"],
+            cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+        )
+        assert r3.returncode == 0, r3.stderr[-3000:]
+        pngs = list(mode_dir.glob("*.png"))
+        gifs = list(mode_dir.glob("*.gif"))
+        assert pngs and gifs, f"{flag}: no outputs in {mode_dir}"
+
 
 @pytest.mark.slow
 def test_bench_contract_cpu(tmp_path):
