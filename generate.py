@@ -186,6 +186,14 @@ def main():
 
     if cfg.channels == 1 and seq.shape[2] == 3:
         seq = seq.mean(dim=2, keepdim=True)
+    if seq.shape[-2:] != (cfg.image_width, cfg.image_width):
+        # arbitrary input resolutions: resize to the model's training size
+        t_, b_, c_ = seq.shape[:3]
+        seq = torch.nn.functional.interpolate(
+            seq.reshape(t_ * b_, c_, *seq.shape[-2:]),
+            size=(cfg.image_width, cfg.image_width),
+            mode="bilinear", align_corners=False,
+        ).reshape(t_, b_, c_, cfg.image_width, cfg.image_width)
     seq = seq.to(device)
     seq_len = len(seq)
 

@@ -63,6 +63,26 @@ def test_train_then_generate(tmp_path):
         assert (out_dir / f"len_{length}-gt.png").exists()
         assert (out_dir / f"len_{length}-gen_full.gif").exists()
 
+    # --start_img/--end_img pair input, off-size frames (auto-resized)
+    import numpy as np
+    from PIL import Image
+
+    pair_dir = tmp_path / "pair"
+    pair_dir.mkdir()
+    for name in ("s.png", "e.png"):
+        arr = (np.random.rand(32, 32, 3) * 255).astype("uint8")
+        Image.fromarray(arr).save(pair_dir / name)
+    pair_out = tmp_path / "gen_pair"
+    rp = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "generate.py"),
+         "--ckpt", str(ckpt), "--start_img", str(pair_dir / "s.png"),
+         "--end_img", str(pair_dir / "e.png"),
+         "--output_root", str(pair_out), "--device", "cpu"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert rp.returncode == 0, rp.stderr[-3000:]
+    assert list(pair_out.glob("*.gif")), "no pair-input outputs"
+
     # --multi_cp and --loop modes (reference README demo flows)
     for flag, stem in (("--multi_cp", "multicpgen"), ("--loop", "loopgen")):
         mode_dir = tmp_path / f"gen_out_{stem}"
