@@ -298,3 +298,26 @@ def test_train_ddp_2rank_gloo_cpu(tmp_path):
     assert (run / "model_0.pth").exists()
     # exactly one rank wrote the checkpoint/log set
     assert (run / "cmd.txt").exists() and (run / "scalars.jsonl").exists()
+
+
+def test_train_h36m_cli_cpu(tmp_path):
+    """h36m branch of the training CLI: skeleton modality (tuple batches),
+    mlp backbone, 3D-visualizer qualitative eval — synthetic fallback data."""
+    env = os.environ.copy()
+    env["PYTHONPATH"] = ROOT
+    log_dir = tmp_path / "logs" / "run"
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "train.py"),
+         "--dataset", "h36m", "--backbone", "mlp", "--batch_size", "2",
+         "--max_seq_len", "6", "--delta_len", "1", "--g_dim", "32",
+         "--z_dim", "4", "--rnn_size", "32", "--nepochs", "1",
+         "--epoch_size", "2", "--nsample", "2", "--device", "cpu",
+         "--qual_iter", "1", "--data_root", "/nonexistent",
+         "--num_workers", "0", "--log_dir", str(log_dir)],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-3000:]
+    run = [d for d in (tmp_path / "logs").iterdir() if d.is_dir()][0]
+    assert (run / "model_0.pth").exists()
+    vis = list((run / "gen_vis").glob("*"))
+    assert vis, "no h36m qualitative outputs written"
