@@ -209,3 +209,58 @@ def test_gauss_stacked_cache_refreshes_on_init_hidden():
     assert ws3 is ws  # persistent buffer, stable storage for hipGraph replay
     assert torch.equal(ws[:10], stack.mu_net.weight)
     assert torch.equal(bs3[:10], stack.mu_net.bias)
+
+
+def test_graph_warmup_snapshot_restore_cpu(tiny_cfg):
+    """_snapshot_train_state/_restore_train_state (the zero-net-update
+    capture-warmup mechanism) restore weights AND optimizer state bitwise,
+    and zero Adam state created after the snapshot."""
+    import torch
+
+    from p2pvg_amd.models import P2PModel
+    from p2pvg_amd.runtime import GraphedTrainStep
+
+    torch.manual_seed(0)
+    model = P2PModel(tiny_cfg)
+    stepper = GraphedTrainStep(model)
+    x = torch.rand(tiny_cfg.max_seq_len, tiny_cfg.batch_size,
+                   tiny_cfg.channels, tiny_cfg.image_width,
+                   tiny_cfg.image_width)
+
+    def one_step():
+        model.zero_grad(set_to_none=False)
+        model(x, 0, len(x) - 1)
+
+    one_step()  # creates Adam state
+    snap = stepper._snapshot_train_state()
+    ref_sd = {k: v.clone() for k, v in model.state_dict().items()}
+
+    one_step()
+    one_step()
+    changed = any(not torch.equal(v, model.state_dict()[k])
+                  for k, v in ref_sd.items() if v.is_floating_point())
+    assert changed, "steps did not change weights; test is vacuous"
+
+    stepper._restore_train_state(snap)
+    for k, v in model.state_dict().items():
+        assert torch.equal(v, ref_sd[k]), f"weight not restored: {k}"
+    for name, opt in stepper._optimizers():
+        msnap = snap[1][name]
+        for p, st in opt.state.items():
+            for kk, vv in st.items():
+                if torch.is_tensor(vv):
+                    assert torch.equal(vv, msnap[id(p)][kk]), \
+                        f"opt state not restored: {name}/{kk}"
+
+    # state created AFTER an early snapshot gets reset to zeros on restore
+    model2 = P2PModel(tiny_cfg)
+    stepper2 = GraphedTrainStep(model2)
+    snap2 = stepper2._snapshot_train_state()  # before any optimizer state
+    model2.zero_grad(set_to_none=False)
+    model2(x, 0, len(x) - 1)
+    stepper2._restore_train_state(snap2)
+    for _, opt in stepper2._optimizers():
+        for _, st in opt.state.items():
+            for kk, vv in st.items():
+                if torch.is_tensor(vv) and vv.is_floating_point():
+                    assert vv.abs().sum() == 0, f"fresh state not zeroed: {kk}"
