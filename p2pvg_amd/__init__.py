@@ -9,3 +9,11 @@ recurrence, RCCL-over-xGMI data parallelism.
 __version__ = "0.1.0"
 
 from .core import Config  # noqa: F401
+
+
+def __getattr__(name):  # lazy top-level exports: keep `import p2pvg_amd` light
+    if name == "P2PModel":
+        from .models import P2PModel
+
+        return P2PModel
+    raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
