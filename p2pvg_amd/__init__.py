@@ -6,7 +6,7 @@ hand-written CDNA4 (gfx950) HIP kernels for the hot ops, hipGraph-captured
 recurrence, RCCL-over-xGMI data parallelism.
 """
 
-__version__ = "0.1.0"
+__version__ = "1.0.0"
 
 from .core import Config  # noqa: F401
 
