@@ -119,3 +119,37 @@ def test_lstm_stack_tuple_input_matches_cat():
         os.environ.pop("P2PVG_KERNELS", None)
     _cmp(mu_a, mu_b, tol=5e-3, name="mu")
     _cmp(lv_a, lv_b, tol=5e-3, name="lv")
+
+
+def test_gauss_stack_cache_follows_weight_updates():
+    """The per-step stacked-weight cache must track parameter changes: after
+    an (optimizer-like) weight update + init_hidden, the fused head computes
+    with the NEW weights (a stale cache would reproduce the old ones)."""
+    import os
+
+    from p2pvg_amd.models.lstm import gaussian_lstm
+
+    torch.manual_seed(5)
+    stack = gaussian_lstm(258, 10, 256, 1, 4).cuda()
+    h = torch.randn(4, 128, device="cuda")
+    g = torch.randn(4, 128, device="cuda")
+    s1 = torch.full((4, 1), 0.5, device="cuda")
+    s2 = torch.full((4, 1), 0.25, device="cuda")
+    stack.init_hidden(4, "cuda")
+    stack((h, g, s1, s2))  # builds and fills the cache
+
+    with torch.no_grad():  # simulate an optimizer step
+        stack.mu_net.weight.mul_(1.5)
+        stack.logvar_net.bias.add_(0.3)
+
+    stack.init_hidden(4, "cuda")  # per-step refresh point
+    _, mu_a, lv_a = stack((h, g, s1, s2))
+
+    stack.init_hidden(4, "cuda")
+    os.environ["P2PVG_KERNELS"] = "torch"
+    try:
+        _, mu_b, lv_b = stack(torch.cat([h, g, s1, s2], 1))
+    finally:
+        os.environ.pop("P2PVG_KERNELS", None)
+    _cmp(mu_a, mu_b, tol=5e-3, name="mu")
+    _cmp(lv_a, lv_b, tol=5e-3, name="lv")
