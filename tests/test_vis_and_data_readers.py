@@ -117,3 +117,41 @@ def test_generate_image_pair(tmp_path):
     )
     assert r.returncode == 0, r.stderr[-2000:]
     assert (tmp_path / "out" / "len_10-gt.png").exists()
+
+
+def test_cp_border_painters_pixel_semantics():
+    """Pixel-level contract of the control-point border painters
+    (reference misc/visualize.py:13-87): orange pad-wide ring on frame 0,
+    red ring on the control-point frame, cp frame repeated past seq_len;
+    interiors untouched; 1-channel inputs promoted to RGB."""
+    import torch
+
+    from p2pvg_amd.utils.vis import (_CP_RGB, _START_RGB, add_gt_cp_border,
+                                     add_samples_cp_border)
+
+    t, b, h, w, pad = 6, 2, 16, 16, 3
+    seq = torch.full((t, b, 1, h, w), 0.5)
+    out = add_gt_cp_border(seq.clone(), seq_len=4, output_len=6, padding=pad)
+    assert out.shape == (t, b, 3, h, w)
+
+    def ring_ok(frame, rgb):
+        for ch in range(3):
+            assert torch.all(frame[ch, :pad, :] == rgb[ch])      # top band
+            assert torch.all(frame[ch, -pad:, :] == rgb[ch])     # bottom
+            assert torch.all(frame[ch, :, :pad] == rgb[ch])      # left
+            assert torch.all(frame[ch, :, -pad:] == rgb[ch])     # right
+        assert torch.all(frame[:, pad:-pad, pad:-pad] == 0.5)    # interior
+
+    ring_ok(out[0, 0], _START_RGB)                # start frame: orange
+    for i in range(3, 6):                         # cp (seq_len-1) + repeats
+        ring_ok(out[i, 0], _CP_RGB)
+    for i in (1, 2):                              # untouched interior frames
+        assert torch.all(out[i] == 0.5)
+
+    samples = torch.full((2, t, b, 1, h, w), 0.5)
+    s = add_samples_cp_border(samples.clone(), seq_len=4, output_len=5,
+                              padding=pad)
+    assert s.shape == (2, t, b, 3, h, w)
+    ring_ok(s[0, 0, 0], _START_RGB)               # first frame orange
+    ring_ok(s[1, 4, 0], _CP_RGB)                  # last GENERATED frame red
+    assert torch.all(s[0, 2] == 0.5)
