@@ -264,3 +264,30 @@ def test_graph_warmup_snapshot_restore_cpu(tiny_cfg):
             for kk, vv in st.items():
                 if torch.is_tensor(vv) and vv.is_floating_point():
                     assert vv.abs().sum() == 0, f"fresh state not zeroed: {kk}"
+
+
+def test_fused_heads_dispatch_policy(monkeypatch):
+    """_fused_heads gate: cuda-only, batch <= 512 (hipBLASLt wins above,
+    profiles/MEASUREMENTS.md note 3), P2PVG_LSTM_HEADS=0 kill switch."""
+    import torch
+
+    from p2pvg_amd.models.lstm import gaussian_lstm
+
+    stack = gaussian_lstm(258, 10, 256, 1, 4)
+    cpu_small = torch.randn(4, 128)
+    assert not stack._fused_heads(cpu_small)          # CPU: never
+    assert not stack._fused_heads((cpu_small,) * 4)   # tuple form too
+
+    class FakeCuda:
+        is_cuda = True
+
+        def __init__(self, rows):
+            self.shape = (rows, 128)
+
+    from p2pvg_amd import ops
+
+    if ops.hip_available():  # CPU containers build the ext; gate if absent
+        assert stack._fused_heads((FakeCuda(512),))
+        assert not stack._fused_heads((FakeCuda(513),))   # large-batch: GEMM heads
+        monkeypatch.setenv("P2PVG_LSTM_HEADS", "0")
+        assert not stack._fused_heads((FakeCuda(4),))
